@@ -1,0 +1,144 @@
+// CDNA4 kernels for the Adasum reduction (reference: the AVX/F16C host
+// kernels in horovod/common/ops/adasum/adasum.h:414-505 — here they are HIP
+// device kernels so the whole VHDD combine runs on-GPU with no host syncs).
+//
+// Stage 1 (adasum_dots_k): per-tensor double-precision dot(a,b), |a|^2, |b|^2
+//   via wave64 shuffle reduction + LDS + device-scope atomicAdd.
+// Stage 2 (adasum_scaledadd_k): a = acoef*a + bcoef*b where
+//   acoef = 1 - dot/(2|a|^2), bcoef = 1 - dot/(2|b|^2), coefficients derived
+//   on-device from stage 1's output (per-tensor isolation inside a fused
+//   buffer, reference adasum.h:361-377).
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
+
+#include "kernels.h"
+
+namespace hvd {
+namespace gpu {
+
+namespace {
+
+template <typename T>
+__device__ __forceinline__ double to_d(T v) {
+  return (double)v;
+}
+template <>
+__device__ __forceinline__ double to_d<__half>(__half v) {
+  return (double)(float)v;
+}
+template <>
+__device__ __forceinline__ double to_d<__hip_bfloat16>(__hip_bfloat16 v) {
+  return (double)(float)v;
+}
+template <typename T>
+__device__ __forceinline__ T from_d(double v) {
+  return (T)v;
+}
+template <>
+__device__ __forceinline__ __half from_d<__half>(double v) {
+  return (__half)(float)v;
+}
+template <>
+__device__ __forceinline__ __hip_bfloat16 from_d<__hip_bfloat16>(double v) {
+  return (__hip_bfloat16)(float)v;
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void adasum_dots_k(AdasumBatchArgs args,
+                                                     double* dots) {
+  int t = blockIdx.y;
+  if (t >= args.count) return;
+  const T* __restrict__ a = (const T*)args.a[t];
+  const T* __restrict__ b = (const T*)args.b[t];
+  const long long n = (long long)args.numel[t];
+  double d = 0, na = 0, nb = 0;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x) {
+    double av = to_d<T>(a[i]);
+    double bv = to_d<T>(b[i]);
+    d += av * bv;
+    na += av * av;
+    nb += bv * bv;
+  }
+  // wave64 reduction
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    d += __shfl_down(d, off, 64);
+    na += __shfl_down(na, off, 64);
+    nb += __shfl_down(nb, off, 64);
+  }
+  __shared__ double lds[3][4];  // 256 threads = 4 waves
+  int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) {
+    lds[0][wave] = d;
+    lds[1][wave] = na;
+    lds[2][wave] = nb;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double bd = lds[0][0] + lds[0][1] + lds[0][2] + lds[0][3];
+    double bna = lds[1][0] + lds[1][1] + lds[1][2] + lds[1][3];
+    double bnb = lds[2][0] + lds[2][1] + lds[2][2] + lds[2][3];
+    atomicAdd(&dots[t * 3 + 0], bd);
+    atomicAdd(&dots[t * 3 + 1], bna);
+    atomicAdd(&dots[t * 3 + 2], bnb);
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void adasum_scaledadd_k(AdasumBatchArgs args,
+                                                          const double* dots) {
+  int t = blockIdx.y;
+  if (t >= args.count) return;
+  T* __restrict__ a = (T*)args.a[t];
+  const T* __restrict__ b = (const T*)args.b[t];
+  const long long n = (long long)args.numel[t];
+  double dot = dots[t * 3 + 0], na = dots[t * 3 + 1], nb = dots[t * 3 + 2];
+  double ac = na > 0 ? 1.0 - dot / (2.0 * na) : 1.0;
+  double bc = nb > 0 ? 1.0 - dot / (2.0 * nb) : 1.0;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long long)gridDim.x * blockDim.x) {
+    a[i] = from_d<T>(ac * to_d<T>(a[i]) + bc * to_d<T>(b[i]));
+  }
+}
+
+template <typename T>
+hipError_t launch_adasum(const AdasumBatchArgs& args, double* dots, bool stage2,
+                         hipStream_t stream) {
+  // stage2 reads dots; stage1 accumulates into them
+  dim3 grid(64, args.count), block(256);
+  if (stage2)
+    adasum_scaledadd_k<T><<<grid, block, 0, stream>>>(args, dots);
+  else
+    adasum_dots_k<T><<<grid, block, 0, stream>>>(args, dots);
+  return hipGetLastError();
+}
+
+template <bool STAGE2>
+hipError_t dispatch(const AdasumBatchArgs& args, int dt, double* dots,
+                    hipStream_t stream) {
+  if (args.count == 0) return hipSuccess;
+  switch (dt) {
+    case DT_F32: return launch_adasum<float>(args, dots, STAGE2, stream);
+    case DT_F64: return launch_adasum<double>(args, dots, STAGE2, stream);
+    case DT_F16: return launch_adasum<__half>(args, dots, STAGE2, stream);
+    case DT_BF16: return launch_adasum<__hip_bfloat16>(args, dots, STAGE2, stream);
+    default: return hipErrorInvalidValue;
+  }
+}
+
+}  // namespace
+
+hipError_t AdasumDotsLaunch(const AdasumBatchArgs& args, int dt, double* dots,
+                            hipStream_t stream) {
+  return dispatch<false>(args, dt, dots, stream);
+}
+
+hipError_t AdasumScaledAddLaunch(const AdasumBatchArgs& args, int dt,
+                                 const double* dots, hipStream_t stream) {
+  return dispatch<true>(args, dt, const_cast<double*>(dots), stream);
+}
+
+}  // namespace gpu
+}  // namespace hvd

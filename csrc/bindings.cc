@@ -1,0 +1,133 @@
+// pybind11 bindings (reference: horovod/torch/mpi_ops_v2.cc N24).
+//
+// One templated entry point per op instead of per-dtype symbol names; the
+// Python layer (horovod_amd/torch/mpi_ops.py) builds the hvd.* surface on
+// top of these.
+#include <torch/extension.h>
+
+#include "core.h"
+#include "gpu.h"
+
+namespace {
+
+using namespace hvd;
+
+ReduceOp OpFromInt(int op) { return (ReduceOp)op; }
+
+py::tuple WaitHandle(int handle) {
+  std::vector<at::Tensor> outputs;
+  at::Tensor extra;
+  int32_t result_int = -1;
+  Status s;
+  {
+    py::gil_scoped_release nogil;
+    s = State().handles.Wait(handle, outputs, extra, &result_int);
+  }
+  if (!s.ok()) {
+    if (s.type == StatusType::ABORTED)
+      throw std::runtime_error("HorovodInternalError: " + s.reason);
+    throw std::runtime_error(s.reason);
+  }
+  py::list outs;
+  for (auto& t : outputs) outs.append(t);
+  py::object extra_obj = py::none();
+  if (extra.defined()) extra_obj = py::cast(extra);
+  return py::make_tuple(outs, extra_obj, result_int);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "horovod_amd core (MI355X-native: TCP controller + RCCL/xGMI)";
+
+  m.def("init",
+        [](int rank, int size, int local_rank, int local_size, int cross_rank,
+           int cross_size, const std::string& addr, int port,
+           int64_t fusion_threshold, double cycle_time_ms, int cache_capacity,
+           double stall_warning_sec, bool timeline) {
+          ControllerConfig cfg;
+          cfg.fusion_threshold_bytes = fusion_threshold;
+          cfg.cycle_time_ms = cycle_time_ms;
+          cfg.cache_capacity = (size_t)cache_capacity;
+          cfg.stall_warning_sec = stall_warning_sec;
+          cfg.timeline_enabled = timeline;
+          py::gil_scoped_release nogil;
+          InitHorovod(rank, size, local_rank, local_size, cross_rank, cross_size,
+                      addr, port, cfg);
+        },
+        py::arg("rank"), py::arg("size"), py::arg("local_rank"),
+        py::arg("local_size"), py::arg("cross_rank"), py::arg("cross_size"),
+        py::arg("addr"), py::arg("port"), py::arg("fusion_threshold"),
+        py::arg("cycle_time_ms"), py::arg("cache_capacity"),
+        py::arg("stall_warning_sec"), py::arg("timeline"));
+
+  m.def("shutdown", [] {
+    py::gil_scoped_release nogil;
+    ShutdownHorovod();
+  });
+  m.def("is_initialized", &IsInitialized);
+
+  m.def("rank", [] { return State().rank; });
+  m.def("size", [] { return State().size; });
+  m.def("local_rank", [] { return State().local_rank; });
+  m.def("local_size", [] { return State().local_size; });
+  m.def("cross_rank", [] { return State().cross_rank; });
+  m.def("cross_size", [] { return State().cross_size; });
+  m.def("rccl_used", [] { return hvd::gpu::RcclUsed(); });
+
+  m.def("set_fusion_threshold",
+        [](int64_t b) { if (State().controller) State().controller->set_fusion_threshold(b); });
+  m.def("get_fusion_threshold",
+        [] { return State().controller ? State().controller->fusion_threshold() : 0; });
+  m.def("set_cycle_time_ms",
+        [](double ms) { if (State().controller) State().controller->set_cycle_time_ms(ms); });
+  m.def("get_cycle_time_ms",
+        [] { return State().controller ? State().controller->cycle_time_ms() : 0.0; });
+
+  // ---- enqueue ops --------------------------------------------------------
+  m.def("allreduce_async",
+        [](std::vector<at::Tensor> tensors, std::vector<at::Tensor> outputs,
+           std::vector<std::string> names, int op, double prescale,
+           double postscale, int process_set, int wire_dtype) {
+          return EnqueueAllreduceMulti(std::move(tensors), std::move(outputs),
+                                       std::move(names), OpFromInt(op), prescale,
+                                       postscale, process_set,
+                                       (DataType)wire_dtype);
+        });
+  m.def("allgather_async", [](at::Tensor t, const std::string& name, int ps) {
+    return EnqueueAllgather(t, name, ps);
+  });
+  m.def("broadcast_async",
+        [](at::Tensor t, at::Tensor out, int root, const std::string& name,
+           int ps) { return EnqueueBroadcast(t, out, root, name, ps); });
+  m.def("alltoall_async",
+        [](at::Tensor t, at::Tensor splits, const std::string& name, int ps) {
+          return EnqueueAlltoall(t, splits, name, ps);
+        });
+  m.def("reducescatter_async",
+        [](at::Tensor t, const std::string& name, int op, double prescale,
+           double postscale, int ps) {
+          return EnqueueReducescatter(t, name, OpFromInt(op), prescale, postscale,
+                                      ps);
+        });
+  m.def("join_async", [](int device, int ps) { return EnqueueJoin(device, ps); });
+  m.def("barrier_async", [](int ps) { return EnqueueBarrier(ps); });
+
+  // ---- completion ---------------------------------------------------------
+  m.def("poll", [](int handle) { return State().handles.Poll(handle); });
+  m.def("wait", &WaitHandle);
+
+  // ---- process sets -------------------------------------------------------
+  m.def("add_process_set", [](std::vector<int32_t> ranks) {
+    if (!State().controller) throw std::runtime_error("not initialized");
+    return State().controller->AddProcessSet(ranks);
+  });
+  m.def("remove_process_set", [](int id) {
+    if (State().controller) State().controller->RemoveProcessSet(id);
+  });
+  m.def("process_set_ranks", [](int id) {
+    if (!State().controller || !State().controller->has_process_set(id))
+      throw std::runtime_error("unknown process set");
+    return State().controller->process_set(id).ranks;
+  });
+}

@@ -1,0 +1,224 @@
+#include "comm.h"
+
+#include <arpa/inet.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <chrono>
+#include <cstring>
+#include <stdexcept>
+#include <thread>
+
+namespace hvd {
+
+namespace {
+
+void set_nodelay(int fd) {
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+}
+
+[[noreturn]] void comm_error(const std::string& what) {
+  throw std::runtime_error("horovod_amd comm: " + what + " (" +
+                           std::strerror(errno) + ")");
+}
+
+}  // namespace
+
+StarComm::~StarComm() { Shutdown(); }
+
+void StarComm::Init(int rank, int size, const std::string& addr, int port,
+                    double timeout_sec) {
+  rank_ = rank;
+  size_ = size;
+  if (size_ <= 1) {
+    alive_ = true;
+    return;
+  }
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::duration<double>(timeout_sec);
+  if (rank_ == 0) {
+    fds_.assign(size_, -1);
+    listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
+    if (listen_fd_ < 0) comm_error("socket");
+    int one = 1;
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in sa{};
+    sa.sin_family = AF_INET;
+    sa.sin_addr.s_addr = INADDR_ANY;
+    sa.sin_port = htons((uint16_t)port);
+    if (bind(listen_fd_, (sockaddr*)&sa, sizeof(sa)) != 0) comm_error("bind");
+    if (listen(listen_fd_, size_) != 0) comm_error("listen");
+    int connected = 0;
+    while (connected < size_ - 1) {
+      int fd = accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) comm_error("accept");
+      set_nodelay(fd);
+      int32_t peer_rank = -1;
+      RecvRaw(fd, &peer_rank, sizeof(peer_rank));
+      if (peer_rank <= 0 || peer_rank >= size_) comm_error("bad peer rank");
+      fds_[peer_rank] = fd;
+      ++connected;
+    }
+  } else {
+    fds_.assign(1, -1);
+    // resolve
+    addrinfo hints{}, *res = nullptr;
+    hints.ai_family = AF_INET;
+    hints.ai_socktype = SOCK_STREAM;
+    std::string port_s = std::to_string(port);
+    if (getaddrinfo(addr.c_str(), port_s.c_str(), &hints, &res) != 0 || !res)
+      comm_error("getaddrinfo " + addr);
+    int fd = -1;
+    while (true) {
+      fd = socket(AF_INET, SOCK_STREAM, 0);
+      if (fd < 0) comm_error("socket");
+      if (connect(fd, res->ai_addr, res->ai_addrlen) == 0) break;
+      close(fd);
+      fd = -1;
+      if (std::chrono::steady_clock::now() > deadline) {
+        freeaddrinfo(res);
+        comm_error("connect timeout to " + addr + ":" + port_s);
+      }
+      std::this_thread::sleep_for(std::chrono::milliseconds(50));
+    }
+    freeaddrinfo(res);
+    set_nodelay(fd);
+    int32_t my_rank = rank_;
+    SendRaw(fd, &my_rank, sizeof(my_rank));
+    fds_[0] = fd;
+  }
+  alive_ = true;
+}
+
+void StarComm::Shutdown() {
+  for (int fd : fds_)
+    if (fd >= 0) close(fd);
+  fds_.clear();
+  if (listen_fd_ >= 0) close(listen_fd_);
+  listen_fd_ = -1;
+  alive_ = false;
+}
+
+int StarComm::FdFor(int peer_rank) const {
+  if (rank_ == 0) return fds_[peer_rank];
+  return fds_[0];
+}
+
+void StarComm::SendRaw(int fd, const void* data, size_t len) {
+  const char* p = (const char*)data;
+  while (len > 0) {
+    ssize_t n = ::send(fd, p, len, MSG_NOSIGNAL);
+    if (n <= 0) comm_error("send");
+    p += n;
+    len -= (size_t)n;
+  }
+}
+
+void StarComm::RecvRaw(int fd, void* data, size_t len) {
+  char* p = (char*)data;
+  while (len > 0) {
+    ssize_t n = ::recv(fd, p, len, 0);
+    if (n <= 0) comm_error("recv");
+    p += n;
+    len -= (size_t)n;
+  }
+}
+
+void StarComm::SendFrame(int fd, const std::string& payload) {
+  uint32_t len = (uint32_t)payload.size();
+  SendRaw(fd, &len, sizeof(len));
+  if (len) SendRaw(fd, payload.data(), len);
+}
+
+std::string StarComm::RecvFrame(int fd) {
+  uint32_t len = 0;
+  RecvRaw(fd, &len, sizeof(len));
+  std::string s(len, '\0');
+  if (len) RecvRaw(fd, &s[0], len);
+  return s;
+}
+
+std::vector<std::string> StarComm::Gather(const std::string& payload) {
+  if (size_ == 1) return {payload};
+  if (rank_ == 0) {
+    std::vector<std::string> out(size_);
+    out[0] = payload;
+    for (int r = 1; r < size_; ++r) out[r] = RecvFrame(fds_[r]);
+    return out;
+  }
+  SendFrame(fds_[0], payload);
+  return {};
+}
+
+std::string StarComm::Bcast(const std::string& payload) {
+  if (size_ == 1) return payload;
+  if (rank_ == 0) {
+    for (int r = 1; r < size_; ++r) SendFrame(fds_[r], payload);
+    return payload;
+  }
+  return RecvFrame(fds_[0]);
+}
+
+void StarComm::Barrier() {
+  Gather("");
+  Bcast("");
+}
+
+void StarComm::BitAnd(std::string& bits) {
+  auto all = Gather(bits);
+  if (rank_ == 0) {
+    std::string acc = all[0];
+    for (int r = 1; r < size_; ++r) {
+      if (all[r].size() != acc.size())
+        throw std::runtime_error("horovod_amd comm: bitvector length mismatch");
+      for (size_t i = 0; i < acc.size(); ++i) acc[i] &= all[r][i];
+    }
+    bits = Bcast(acc);
+  } else {
+    bits = Bcast("");
+  }
+}
+
+void StarComm::BitOr(std::string& bits) {
+  auto all = Gather(bits);
+  if (rank_ == 0) {
+    std::string acc = all[0];
+    for (int r = 1; r < size_; ++r) {
+      if (all[r].size() != acc.size())
+        throw std::runtime_error("horovod_amd comm: bitvector length mismatch");
+      for (size_t i = 0; i < acc.size(); ++i) acc[i] |= all[r][i];
+    }
+    bits = Bcast(acc);
+  } else {
+    bits = Bcast("");
+  }
+}
+
+std::string StarComm::ScatterFrames(const std::vector<std::string>& frames) {
+  if (size_ == 1) return frames.empty() ? std::string() : frames[0];
+  if (rank_ == 0) {
+    for (int r = 1; r < size_; ++r)
+      SendFrame(fds_[r], r < (int)frames.size() ? frames[r] : std::string());
+    return frames.empty() ? std::string() : frames[0];
+  }
+  return RecvFrame(fds_[0]);
+}
+
+void StarComm::SendToRank(int dst, const void* data, size_t len) {
+  SendRaw(fds_[dst], data, len);
+}
+void StarComm::RecvFromRank(int src, void* data, size_t len) {
+  RecvRaw(fds_[src], data, len);
+}
+void StarComm::SendToRoot(const void* data, size_t len) {
+  SendRaw(fds_[0], data, len);
+}
+void StarComm::RecvFromRoot(void* data, size_t len) {
+  RecvRaw(fds_[0], data, len);
+}
+
+}  // namespace hvd

@@ -1,0 +1,825 @@
+#include "core.h"
+
+#include <chrono>
+#include <cstdio>
+#include <cstring>
+#include <stdexcept>
+
+#include "gpu.h"
+#include "timeline.h"
+
+namespace hvd {
+
+namespace {
+std::string SetKey(int32_t set_id, const std::string& name) {
+  return std::to_string(set_id) + ":" + name;
+}
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// TensorQueue
+// ---------------------------------------------------------------------------
+Status TensorQueue::Add(Request req, TensorTableEntry entry) {
+  std::lock_guard<std::mutex> g(mu_);
+  std::string key = SetKey(req.process_set_id, req.name);
+  if (table_.count(key)) {
+    return Status::InvalidArgument(
+        "Duplicate tensor name in queue: " + req.name +
+        ". This happens if multiple outstanding collectives share a name; pass "
+        "a unique name= to each call.");
+  }
+  table_.emplace(std::move(key), std::move(entry));
+  messages_.push_back(std::move(req));
+  return Status::OK();
+}
+
+Status TensorQueue::AddMulti(std::vector<Request>& reqs,
+                             std::vector<TensorTableEntry>& entries) {
+  std::lock_guard<std::mutex> g(mu_);
+  for (auto& r : reqs) {
+    if (table_.count(SetKey(r.process_set_id, r.name)))
+      return Status::InvalidArgument("Duplicate tensor name in queue: " + r.name);
+  }
+  for (size_t i = 0; i < reqs.size(); ++i) {
+    table_.emplace(SetKey(reqs[i].process_set_id, reqs[i].name),
+                   std::move(entries[i]));
+    messages_.push_back(std::move(reqs[i]));
+  }
+  return Status::OK();
+}
+
+std::vector<Request> TensorQueue::PopMessages() {
+  std::lock_guard<std::mutex> g(mu_);
+  std::vector<Request> out(messages_.begin(), messages_.end());
+  messages_.clear();
+  return out;
+}
+
+bool TensorQueue::PopEntry(int32_t set_id, const std::string& name,
+                           TensorTableEntry& out) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = table_.find(SetKey(set_id, name));
+  if (it == table_.end()) return false;
+  out = std::move(it->second);
+  table_.erase(it);
+  return true;
+}
+
+void TensorQueue::FailAll(const Status& s) {
+  std::lock_guard<std::mutex> g(mu_);
+  for (auto& kv : table_) {
+    if (kv.second.callback) kv.second.callback(s, kv.second);
+  }
+  table_.clear();
+  messages_.clear();
+}
+
+size_t TensorQueue::size() const {
+  std::lock_guard<std::mutex> g(mu_);
+  return table_.size();
+}
+
+// ---------------------------------------------------------------------------
+// HandleManager
+// ---------------------------------------------------------------------------
+int HandleManager::Allocate(int n_outputs) {
+  std::lock_guard<std::mutex> g(mu_);
+  int h = next_++;
+  auto hs = std::make_shared<HandleState>();
+  hs->outputs.resize(n_outputs);
+  handles_[h] = hs;
+  return h;
+}
+
+std::shared_ptr<HandleState> HandleManager::Get(int handle) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = handles_.find(handle);
+  return it == handles_.end() ? nullptr : it->second;
+}
+
+bool HandleManager::Poll(int handle) {
+  auto h = Get(handle);
+  if (!h) return true;
+  std::lock_guard<std::mutex> g(h->mu);
+  return h->done;
+}
+
+Status HandleManager::Wait(int handle, std::vector<at::Tensor>& outputs,
+                           at::Tensor& extra, int32_t* result_int) {
+  auto h = Get(handle);
+  if (!h) return Status::InvalidArgument("unknown handle");
+  std::unique_lock<std::mutex> lk(h->mu);
+  h->cv.wait(lk, [&] { return h->done; });
+  outputs = h->outputs;
+  extra = h->extra;
+  if (result_int) *result_int = h->result_int;
+  Status s = h->status;
+  lk.unlock();
+  Drop(handle);
+  return s;
+}
+
+void HandleManager::MarkDone(int handle, const Status& s) {
+  auto h = Get(handle);
+  if (!h) return;
+  std::lock_guard<std::mutex> g(h->mu);
+  h->status = s;
+  h->done = true;
+  h->cv.notify_all();
+}
+
+void HandleManager::FailAll(const Status& s) {
+  std::lock_guard<std::mutex> g(mu_);
+  for (auto& kv : handles_) {
+    std::lock_guard<std::mutex> g2(kv.second->mu);
+    if (!kv.second->done) {
+      kv.second->status = s;
+      kv.second->done = true;
+      kv.second->cv.notify_all();
+    }
+  }
+}
+
+void HandleManager::Drop(int handle) {
+  std::lock_guard<std::mutex> g(mu_);
+  handles_.erase(handle);
+}
+
+// ---------------------------------------------------------------------------
+// Global state
+// ---------------------------------------------------------------------------
+GlobalState& State() {
+  static GlobalState st;
+  return st;
+}
+
+bool IsInitialized() { return State().initialized; }
+
+namespace {
+
+void FailEntries(std::vector<TensorTableEntry>& entries, const Status& s) {
+  for (auto& e : entries)
+    if (e.callback) e.callback(s, e);
+}
+
+std::vector<std::vector<int64_t>> ParseShapes(const Response& r) {
+  std::vector<std::vector<int64_t>> shapes;
+  const int64_t* p = r.tensor_shapes.data();
+  const int64_t* end = p + r.tensor_shapes.size();
+  while (p < end) {
+    int64_t nd = *p++;
+    shapes.emplace_back(p, p + nd);
+    p += nd;
+  }
+  return shapes;
+}
+
+at::Tensor FlatPrescaled(const TensorTableEntry& e, DataType wire) {
+  at::Tensor t = e.tensor.flatten();
+  auto wire_t = DataTypeToTorch(wire);
+  if (e.prescale != 1.0) {
+    t = t.to(at::kDouble).mul_(e.prescale).to(wire_t);
+  } else if (t.scalar_type() != wire_t) {
+    t = t.to(wire_t);
+  } else if (!t.is_contiguous()) {
+    t = t.contiguous();
+  }
+  return t;
+}
+
+void UnpackInto(TensorTableEntry& e, at::Tensor flat_slice) {
+  at::Tensor src = flat_slice;
+  if (e.postscale != 1.0) src = src.to(at::kDouble).mul_(e.postscale);
+  if (!e.output.defined()) e.output = at::empty_like(e.tensor);
+  e.output.flatten().copy_(src.reshape(e.output.sizes()).flatten());
+}
+
+// Adasum pairwise combine: a = a*(1 - dot/(2|a|^2)) + b*(1 - dot/(2|b|^2)),
+// per tensor within the fused buffer (reference: adasum.h:347-412
+// FusedPairwiseReduceWithComm — per-tensor coefficient isolation).
+void AdasumCombineInPlace(at::Tensor a, at::Tensor b,
+                          const std::vector<int64_t>& offsets,
+                          const std::vector<int64_t>& counts) {
+  for (size_t i = 0; i < offsets.size(); ++i) {
+    auto av = a.narrow(0, offsets[i], counts[i]);
+    auto bv = b.narrow(0, offsets[i], counts[i]);
+    double dot = av.dot(bv).item<double>();
+    double na = av.dot(av).item<double>();
+    double nb = bv.dot(bv).item<double>();
+    double acoef = na > 0 ? 1.0 - dot / (2.0 * na) : 1.0;
+    double bcoef = nb > 0 ? 1.0 - dot / (2.0 * nb) : 1.0;
+    av.mul_(acoef).add_(bv, bcoef);
+  }
+}
+
+// --------------------------- CPU data plane --------------------------------
+// All CPU collectives are star-routed through global rank 0 and are GLOBAL
+// lock-step frames: every rank (member or not) participates in exactly one
+// Gather and one Bcast (plus ScatterFrames for alltoall), so control-plane
+// framing can never interleave with data.  This is the correctness tier —
+// the performance tier is RCCL over xGMI (gpu.cc).
+
+void PerTensorLayout(const Response& resp, std::vector<int64_t>& offsets,
+                     std::vector<int64_t>& counts) {
+  auto shapes = ParseShapes(resp);
+  int64_t off = 0;
+  for (auto& sh : shapes) {
+    int64_t n = 1;
+    for (auto d : sh) n *= d;
+    offsets.push_back(off);
+    counts.push_back(n);
+    off += n;
+  }
+}
+
+void CPUAllreduce(GlobalState& st, const Response& resp,
+                  std::vector<TensorTableEntry>& entries, bool member) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  DataType wire = resp.dtype;
+  auto wire_t = DataTypeToTorch(wire);
+  std::string payload;
+  std::vector<int64_t> offsets, counts;
+  PerTensorLayout(resp, offsets, counts);
+  if (member && !entries.empty()) {
+    std::vector<at::Tensor> flats;
+    for (auto& e : entries) flats.push_back(FlatPrescaled(e, wire));
+    at::Tensor buf = flats.size() == 1 ? flats[0] : at::cat(flats);
+    payload.assign((const char*)buf.data_ptr(), buf.numel() * buf.element_size());
+  }
+  auto gathered = st.comm.Gather(payload);
+  std::string result;
+  if (st.comm.is_root()) {
+    std::vector<at::Tensor> bufs;
+    for (int r : set.ranks) {
+      if (r < (int)gathered.size() && !gathered[r].empty()) {
+        auto& s = gathered[r];
+        int64_t n = (int64_t)(s.size() / DataTypeSize(wire));
+        bufs.push_back(at::from_blob((void*)s.data(), {n}, wire_t).clone());
+      }
+    }
+    if (!bufs.empty()) {
+      at::Tensor acc;
+      if (resp.type == ResponseType::ADASUM) {
+        std::vector<at::Tensor> work;
+        for (auto& b : bufs) work.push_back(b.to(at::kDouble));
+        size_t p = 1;
+        while (p * 2 <= work.size()) p *= 2;
+        for (size_t i = p; i < work.size(); ++i)
+          AdasumCombineInPlace(work[i - p], work[i], offsets, counts);
+        work.resize(p);
+        for (size_t stride = 1; stride < p; stride *= 2)
+          for (size_t i = 0; i + stride < p; i += 2 * stride)
+            AdasumCombineInPlace(work[i], work[i + stride], offsets, counts);
+        acc = work[0].to(wire_t);
+      } else {
+        acc = bufs[0];
+        for (size_t i = 1; i < bufs.size(); ++i) {
+          switch (resp.reduce_op) {
+            case ReduceOp::MIN: acc = at::minimum(acc, bufs[i]); break;
+            case ReduceOp::MAX: acc = at::maximum(acc, bufs[i]); break;
+            case ReduceOp::PRODUCT: acc = acc.mul(bufs[i]); break;
+            default: acc = acc.add(bufs[i]); break;
+          }
+        }
+      }
+      result.assign((const char*)acc.data_ptr(), acc.numel() * acc.element_size());
+    }
+  }
+  result = st.comm.Bcast(result);
+  if (member && !entries.empty()) {
+    int64_t n = (int64_t)(result.size() / DataTypeSize(wire));
+    at::Tensor buf = at::from_blob((void*)result.data(), {n}, wire_t);
+    for (size_t i = 0; i < entries.size(); ++i)
+      UnpackInto(entries[i], buf.narrow(0, offsets[i], counts[i]));
+  }
+}
+
+void CPUAllgather(GlobalState& st, const Response& resp,
+                  std::vector<TensorTableEntry>& entries, bool member) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  std::string payload;
+  if (member && !entries.empty()) {
+    at::Tensor t = entries[0].tensor.contiguous();
+    payload.assign((const char*)t.data_ptr(), t.numel() * t.element_size());
+  }
+  auto gathered = st.comm.Gather(payload);
+  std::string result;
+  if (st.comm.is_root()) {
+    for (int r : set.ranks)
+      if (r < (int)gathered.size()) result += gathered[r];
+  }
+  result = st.comm.Bcast(result);
+  if (member && !entries.empty()) {
+    auto& e = entries[0];
+    auto shapes = ParseShapes(resp);
+    std::vector<int64_t> out_shape = shapes[0];
+    int64_t total0 = 0;
+    for (auto s : resp.tensor_sizes) total0 += s;
+    if (out_shape.empty()) out_shape = {total0};
+    else out_shape[0] = total0;
+    e.output = at::empty(out_shape, e.tensor.options());
+    std::memcpy(e.output.data_ptr(), result.data(),
+                std::min((size_t)(e.output.numel() * e.output.element_size()),
+                         result.size()));
+  }
+}
+
+void CPUBroadcast(GlobalState& st, const Response& resp,
+                  std::vector<TensorTableEntry>& entries, bool member) {
+  std::string payload;
+  if (member && !entries.empty() && resp.root_rank == st.rank) {
+    at::Tensor t = entries[0].tensor.contiguous();
+    payload.assign((const char*)t.data_ptr(), t.numel() * t.element_size());
+  }
+  auto gathered = st.comm.Gather(payload);
+  std::string result;
+  if (st.comm.is_root() && resp.root_rank < (int)st.size &&
+      resp.root_rank < (int)gathered.size())
+    result = gathered[resp.root_rank];
+  result = st.comm.Bcast(result);
+  if (member && !entries.empty()) {
+    auto& e = entries[0];
+    if (!e.output.defined()) e.output = at::empty_like(e.tensor);
+    if ((size_t)(e.output.numel() * e.output.element_size()) == result.size()) {
+      if (e.output.is_contiguous()) {
+        std::memcpy(e.output.data_ptr(), result.data(), result.size());
+      } else {
+        int64_t n = e.output.numel();
+        at::Tensor src = at::from_blob((void*)result.data(), {n},
+                                       e.output.options().device(at::kCPU));
+        e.output.copy_(src.reshape(e.output.sizes()));
+      }
+    }
+  }
+}
+
+void CPUAlltoall(GlobalState& st, const Response& resp,
+                 std::vector<TensorTableEntry>& entries, bool member) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  int n = (int)set.ranks.size();
+  std::string payload;
+  if (member && !entries.empty()) {
+    at::Tensor t = entries[0].tensor.contiguous();
+    payload.assign((const char*)t.data_ptr(), t.numel() * t.element_size());
+  }
+  auto gathered = st.comm.Gather(payload);
+  std::vector<std::string> outgoing((size_t)st.size);
+  if (st.comm.is_root()) {
+    auto shapes = ParseShapes(resp);
+    int64_t row_bytes = (int64_t)DataTypeSize(resp.dtype);
+    if (!shapes.empty())
+      for (size_t d = 1; d < shapes[0].size(); ++d) row_bytes *= shapes[0][d];
+    for (int i = 0; i < n; ++i) {
+      int src_global = set.ranks[i];
+      if (src_global >= (int)gathered.size()) continue;
+      const std::string& buf = gathered[src_global];
+      int64_t off = 0;
+      for (int j = 0; j < n; ++j) {
+        int64_t rows = resp.tensor_sizes[(size_t)i * n + j];
+        int dst_global = set.ranks[j];
+        outgoing[dst_global].append(buf.data() + off * row_bytes,
+                                    (size_t)(rows * row_bytes));
+        off += rows;
+      }
+    }
+  }
+  std::string mine = st.comm.ScatterFrames(outgoing);
+  if (member && !entries.empty()) {
+    auto& e = entries[0];
+    int li = set.local_index(st.rank);
+    int64_t total_rows = 0;
+    for (int i = 0; i < n; ++i) total_rows += resp.tensor_sizes[(size_t)i * n + li];
+    std::vector<int64_t> out_shape(e.tensor.sizes().begin(), e.tensor.sizes().end());
+    if (out_shape.empty()) out_shape = {total_rows};
+    else out_shape[0] = total_rows;
+    e.output = at::empty(out_shape, e.tensor.options());
+    std::memcpy(e.output.data_ptr(), mine.data(),
+                std::min((size_t)(e.output.numel() * e.output.element_size()),
+                         mine.size()));
+    e.received_splits = at::empty({n}, at::kLong);
+    auto* rs = e.received_splits.data_ptr<int64_t>();
+    for (int i = 0; i < n; ++i) rs[i] = resp.tensor_sizes[(size_t)i * n + li];
+  }
+}
+
+void CPUReducescatter(GlobalState& st, const Response& resp,
+                      std::vector<TensorTableEntry>& entries, bool member) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  int n = (int)set.ranks.size();
+  DataType wire = resp.dtype;
+  auto wire_t = DataTypeToTorch(wire);
+  std::string payload;
+  if (member && !entries.empty()) {
+    auto f = FlatPrescaled(entries[0], wire);
+    payload.assign((const char*)f.data_ptr(), f.numel() * f.element_size());
+  }
+  auto gathered = st.comm.Gather(payload);
+  std::string result;
+  if (st.comm.is_root()) {
+    at::Tensor acc;
+    for (int r : set.ranks) {
+      if (r >= (int)gathered.size() || gathered[r].empty()) continue;
+      int64_t cnt = (int64_t)(gathered[r].size() / DataTypeSize(wire));
+      auto t = at::from_blob((void*)gathered[r].data(), {cnt}, wire_t).clone();
+      acc = acc.defined() ? acc.add(t) : t;
+    }
+    if (acc.defined())
+      result.assign((const char*)acc.data_ptr(), acc.numel() * acc.element_size());
+  }
+  result = st.comm.Bcast(result);
+  if (member && !entries.empty()) {
+    auto& e = entries[0];
+    auto shapes = ParseShapes(resp);
+    std::vector<int64_t> shape = shapes[0];
+    int64_t first = shape.empty() ? 1 : shape[0];
+    int64_t base = first / n, rem = first % n;
+    int li = set.local_index(st.rank);
+    int64_t my_rows = base + (li < rem ? 1 : 0);
+    int64_t row_off = li * base + std::min<int64_t>(li, rem);
+    int64_t row_elems = 1;
+    for (size_t d = 1; d < shape.size(); ++d) row_elems *= shape[d];
+    std::vector<int64_t> out_shape = shape;
+    if (out_shape.empty()) out_shape = {my_rows};
+    else out_shape[0] = my_rows;
+    int64_t cnt = (int64_t)(result.size() / DataTypeSize(wire));
+    at::Tensor full = at::from_blob((void*)result.data(), {cnt}, wire_t);
+    e.output = at::empty(out_shape, e.tensor.options());
+    at::Tensor src = full.narrow(0, row_off * row_elems, my_rows * row_elems);
+    if (e.postscale != 1.0) src = src.to(at::kDouble).mul_(e.postscale);
+    e.output.flatten().copy_(src.to(e.output.scalar_type()));
+  }
+}
+
+// ------------------------- dispatch ----------------------------------------
+
+void PerformOperation(GlobalState& st, Response& resp) {
+  if (!st.controller->has_process_set(resp.process_set_id)) return;
+  auto& set = st.controller->process_set(resp.process_set_id);
+  bool member = set.contains(st.rank);
+  bool joined;
+  {
+    std::lock_guard<std::mutex> g(st.join_mu);
+    joined = st.local_joined.count(resp.process_set_id) &&
+             st.local_joined[resp.process_set_id];
+  }
+
+  if (resp.type == ResponseType::JOIN) {
+    TensorTableEntry e;
+    if (st.queue.PopEntry(resp.process_set_id, "join", e)) {
+      e.join_result = resp.last_joined_rank;
+      if (e.callback) e.callback(Status::OK(), e);
+    }
+    std::lock_guard<std::mutex> g(st.join_mu);
+    st.local_joined[resp.process_set_id] = false;
+    return;
+  }
+  if (resp.type == ResponseType::BARRIER) {
+    TensorTableEntry e;
+    if (st.queue.PopEntry(resp.process_set_id, "barrier", e) && e.callback)
+      e.callback(Status::OK(), e);
+    return;
+  }
+
+  std::vector<TensorTableEntry> entries;
+  if (member && !joined) {
+    for (auto& name : resp.names) {
+      TensorTableEntry e;
+      if (st.queue.PopEntry(resp.process_set_id, name, e)) {
+        entries.push_back(std::move(e));
+      } else if (resp.type != ResponseType::BARRIER) {
+        std::fprintf(stderr, "[horovod_amd] missing entry for %s\n", name.c_str());
+      }
+    }
+  } else if (member && joined &&
+             (resp.type == ResponseType::ALLREDUCE ||
+              resp.type == ResponseType::ADASUM)) {
+    // Zero substitution for a joined rank (reference: tensor_queue.cc
+    // GetTensorEntriesFromResponse join path).
+    auto shapes = ParseShapes(resp);
+    int dev;
+    {
+      std::lock_guard<std::mutex> g(st.join_mu);
+      dev = st.join_device.count(resp.process_set_id)
+                ? st.join_device[resp.process_set_id]
+                : CPU_DEVICE_ID;
+    }
+    for (size_t i = 0; i < resp.names.size(); ++i) {
+      TensorTableEntry e;
+      e.name = resp.names[i];
+      auto opts = at::TensorOptions()
+                      .dtype(DataTypeToTorch(resp.dtype))
+                      .device(dev == CPU_DEVICE_ID ? at::Device(at::kCPU)
+                                                   : at::Device(at::kCUDA, dev));
+      e.tensor = at::zeros(shapes[i], opts);
+      e.device = dev;
+      e.process_set_id = resp.process_set_id;
+      entries.push_back(std::move(e));
+    }
+  }
+
+  if (resp.type == ResponseType::ERROR) {
+    FailEntries(entries, Status::PreconditionError(resp.error_msg));
+    return;
+  }
+
+  if (st.timeline) st.timeline->OpStart(resp);
+
+  // Response device is the normalized marker (-1 CPU, -2 GPU) — the GPU path
+  // must run on every rank (even relay-only non-members) so the RCCL comm
+  // bootstrap's star frames stay lock-step.
+  bool gpu_op = resp.device != CPU_DEVICE_ID;
+  try {
+    if (gpu_op) {
+      gpu::Execute(st, resp, entries);  // async; completion via finalizer
+    } else {
+      switch (resp.type) {
+        case ResponseType::ALLREDUCE:
+        case ResponseType::ADASUM:
+          CPUAllreduce(st, resp, entries, member);
+          break;
+        case ResponseType::ALLGATHER:
+          CPUAllgather(st, resp, entries, member);
+          break;
+        case ResponseType::BROADCAST:
+          CPUBroadcast(st, resp, entries, member);
+          break;
+        case ResponseType::ALLTOALL:
+          CPUAlltoall(st, resp, entries, member);
+          break;
+        case ResponseType::REDUCESCATTER:
+          CPUReducescatter(st, resp, entries, member);
+          break;
+        default:
+          break;
+      }
+      FailEntries(entries, Status::OK());
+      if (st.timeline) st.timeline->OpEnd(resp);
+    }
+  } catch (const std::exception& ex) {
+    FailEntries(entries, Status::UnknownError(ex.what()));
+    if (st.timeline) st.timeline->OpEnd(resp);
+  }
+}
+
+void Abort(GlobalState& st, const std::string& why) {
+  st.aborted = true;
+  st.abort_reason = why;
+  std::fprintf(stderr, "[horovod_amd] background loop aborted: %s\n", why.c_str());
+  auto s = Status::Aborted(why);
+  st.queue.FailAll(s);
+  st.handles.FailAll(s);
+}
+
+void BackgroundLoop(GlobalState& st) {
+  while (true) {
+    auto cycle_start = std::chrono::steady_clock::now();
+    auto reqs = st.queue.PopMessages();
+    ResponseList rl;
+    try {
+      rl = st.controller->RunCycle(std::move(reqs), st.shutdown_requested);
+    } catch (const std::exception& ex) {
+      Abort(st, ex.what());
+      break;
+    }
+    for (auto& resp : rl.responses) PerformOperation(st, resp);
+    if (rl.shutdown) break;
+    if (rl.responses.empty()) {
+      auto elapsed = std::chrono::steady_clock::now() - cycle_start;
+      auto target =
+          std::chrono::duration<double, std::milli>(st.controller->cycle_time_ms());
+      if (elapsed < target) std::this_thread::sleep_for(target - elapsed);
+    }
+  }
+  gpu::WaitAllPending();
+  st.shutting_down = true;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Lifecycle
+// ---------------------------------------------------------------------------
+void InitHorovod(int rank, int size, int local_rank, int local_size,
+                 int cross_rank, int cross_size, const std::string& addr,
+                 int port, const ControllerConfig& cfg) {
+  auto& st = State();
+  if (st.initialized) return;
+  st.rank = rank;
+  st.size = size;
+  st.local_rank = local_rank;
+  st.local_size = local_size;
+  st.cross_rank = cross_rank;
+  st.cross_size = cross_size;
+  st.aborted = false;
+  st.shutdown_requested = false;
+  st.shutting_down = false;
+  st.abort_reason.clear();
+  st.comm.Init(rank, size, addr, port);
+  st.controller.reset(new Controller(&st.comm, rank, size, cfg));
+  if (cfg.timeline_enabled) {
+    const char* tf = std::getenv("HOROVOD_TIMELINE");
+    if (tf) st.timeline.reset(new Timeline(tf, rank));
+  }
+  st.bg_thread = std::thread([&st] { BackgroundLoop(st); });
+  st.initialized = true;
+}
+
+void ShutdownHorovod() {
+  auto& st = State();
+  if (!st.initialized) return;
+  st.shutdown_requested = true;
+  if (st.bg_thread.joinable()) st.bg_thread.join();
+  gpu::Shutdown();
+  st.comm.Shutdown();
+  st.controller.reset();
+  st.timeline.reset();
+  st.local_joined.clear();
+  st.join_device.clear();
+  st.initialized = false;
+}
+
+// ---------------------------------------------------------------------------
+// Enqueue API
+// ---------------------------------------------------------------------------
+namespace {
+
+Request MakeRequest(RequestType type, const std::string& name,
+                    const at::Tensor& t, ReduceOp op, double pre, double post,
+                    int root, int32_t set_id, DataType wire_dtype) {
+  auto& st = State();
+  Request r;
+  r.type = type;
+  r.rank = st.rank;
+  r.name = name;
+  r.dtype = wire_dtype;
+  r.shape.assign(t.sizes().begin(), t.sizes().end());
+  r.root_rank = root;
+  r.reduce_op = op;
+  r.prescale = pre;
+  r.postscale = post;
+  r.process_set_id = set_id;
+  // Normalized device marker: -1 CPU, -2 GPU (executor uses the entry's real
+  // local device index; see controller.h design notes).
+  r.device = t.is_cpu() ? CPU_DEVICE_ID : -2;
+  return r;
+}
+
+TensorTableEntry MakeEntry(const std::string& name, at::Tensor t, at::Tensor out,
+                           ReduceOp op, double pre, double post, int root,
+                           int32_t set_id) {
+  TensorTableEntry e;
+  e.name = name;
+  e.tensor = t;
+  e.output = out;
+  e.device = t.is_cpu() ? CPU_DEVICE_ID : (int)t.get_device();
+  e.root_rank = root;
+  e.reduce_op = op;
+  e.prescale = pre;
+  e.postscale = post;
+  e.process_set_id = set_id;
+  if (e.device != CPU_DEVICE_ID) e.ready_event = gpu::RecordReadyEvent(e.device);
+  return e;
+}
+
+int EnqueueImpl(std::vector<Request> reqs, std::vector<TensorTableEntry> entries) {
+  auto& st = State();
+  if (!st.initialized)
+    throw std::runtime_error(
+        "horovod_amd has not been initialized; call hvd.init() first.");
+  if (st.aborted)
+    throw std::runtime_error("HorovodInternalError: " + st.abort_reason);
+  int handle = st.handles.Allocate((int)entries.size());
+  auto hs = st.handles.Get(handle);
+  auto remaining = std::make_shared<std::atomic<int>>((int)entries.size());
+  for (size_t i = 0; i < entries.size(); ++i) {
+    entries[i].callback = [hs, remaining, i](const Status& s, TensorTableEntry& e) {
+      {
+        std::lock_guard<std::mutex> g(hs->mu);
+        if (hs->outputs.size() <= i) hs->outputs.resize(i + 1);
+        hs->outputs[i] = e.output.defined() ? e.output : e.tensor;
+        if (e.received_splits.defined()) hs->extra = e.received_splits;
+        if (e.join_result >= 0) hs->result_int = e.join_result;
+        if (!s.ok() && hs->status.ok()) hs->status = s;
+      }
+      if (remaining->fetch_sub(1) == 1) {
+        std::lock_guard<std::mutex> g(hs->mu);
+        hs->done = true;
+        hs->cv.notify_all();
+      }
+    };
+  }
+  Status added = st.queue.AddMulti(reqs, entries);
+  if (!added.ok()) st.handles.MarkDone(handle, added);
+  return handle;
+}
+
+}  // namespace
+
+int EnqueueAllreduceMulti(std::vector<at::Tensor> tensors,
+                          std::vector<at::Tensor> outputs,
+                          std::vector<std::string> names, ReduceOp op,
+                          double prescale, double postscale,
+                          int32_t process_set_id, DataType wire_dtype) {
+  std::vector<Request> reqs;
+  std::vector<TensorTableEntry> entries;
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    reqs.push_back(MakeRequest(RequestType::ALLREDUCE, names[i], tensors[i], op,
+                               prescale, postscale, -1, process_set_id, wire_dtype));
+    entries.push_back(MakeEntry(names[i], tensors[i], outputs[i], op, prescale,
+                                postscale, -1, process_set_id));
+  }
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueAllgather(at::Tensor tensor, const std::string& name,
+                     int32_t process_set_id) {
+  auto req = MakeRequest(RequestType::ALLGATHER, name, tensor, ReduceOp::SUM, 1.0,
+                         1.0, -1, process_set_id, DataTypeFromTorch(tensor.scalar_type()));
+  auto e = MakeEntry(name, tensor, at::Tensor(), ReduceOp::SUM, 1.0, 1.0, -1,
+                     process_set_id);
+  std::vector<Request> reqs{std::move(req)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueBroadcast(at::Tensor tensor, at::Tensor output, int root_rank,
+                     const std::string& name, int32_t process_set_id) {
+  auto req = MakeRequest(RequestType::BROADCAST, name, tensor, ReduceOp::SUM, 1.0,
+                         1.0, root_rank, process_set_id,
+                         DataTypeFromTorch(tensor.scalar_type()));
+  auto e = MakeEntry(name, tensor, output, ReduceOp::SUM, 1.0, 1.0, root_rank,
+                     process_set_id);
+  std::vector<Request> reqs{std::move(req)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueAlltoall(at::Tensor tensor, at::Tensor splits, const std::string& name,
+                    int32_t process_set_id) {
+  auto req = MakeRequest(RequestType::ALLTOALL, name, tensor, ReduceOp::SUM, 1.0,
+                         1.0, -1, process_set_id,
+                         DataTypeFromTorch(tensor.scalar_type()));
+  auto e = MakeEntry(name, tensor, at::Tensor(), ReduceOp::SUM, 1.0, 1.0, -1,
+                     process_set_id);
+  if (splits.defined() && splits.numel() > 0) {
+    auto s = splits.to(at::kLong).contiguous();
+    auto* p = s.data_ptr<int64_t>();
+    req.splits.assign(p, p + s.numel());
+    e.splits = req.splits;
+  }
+  std::vector<Request> reqs{std::move(req)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueReducescatter(at::Tensor tensor, const std::string& name, ReduceOp op,
+                         double prescale, double postscale,
+                         int32_t process_set_id) {
+  auto req = MakeRequest(RequestType::REDUCESCATTER, name, tensor, op, prescale,
+                         postscale, -1, process_set_id,
+                         DataTypeFromTorch(tensor.scalar_type()));
+  auto e = MakeEntry(name, tensor, at::Tensor(), op, prescale, postscale, -1,
+                     process_set_id);
+  std::vector<Request> reqs{std::move(req)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueJoin(int device, int32_t process_set_id) {
+  auto& st = State();
+  {
+    std::lock_guard<std::mutex> g(st.join_mu);
+    st.local_joined[process_set_id] = true;
+    st.join_device[process_set_id] = device;
+  }
+  Request r;
+  r.type = RequestType::JOIN;
+  r.rank = st.rank;
+  r.name = "join";
+  r.process_set_id = process_set_id;
+  r.device = device == CPU_DEVICE_ID ? CPU_DEVICE_ID : -2;
+  TensorTableEntry e;
+  e.name = "join";
+  e.device = device;
+  e.process_set_id = process_set_id;
+  std::vector<Request> reqs{std::move(r)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+int EnqueueBarrier(int32_t process_set_id) {
+  auto& st = State();
+  Request r;
+  r.type = RequestType::BARRIER;
+  r.rank = st.rank;
+  r.name = "barrier";
+  r.process_set_id = process_set_id;
+  TensorTableEntry e;
+  e.name = "barrier";
+  e.process_set_id = process_set_id;
+  std::vector<Request> reqs{std::move(r)};
+  std::vector<TensorTableEntry> entries{std::move(e)};
+  return EnqueueImpl(std::move(reqs), std::move(entries));
+}
+
+}  // namespace hvd

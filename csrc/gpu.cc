@@ -1,0 +1,630 @@
+#include "gpu.h"
+
+#include <hip/hip_runtime_api.h>
+#include <rccl/rccl.h>
+
+#include <c10/hip/HIPCachingAllocator.h>
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
+#include <condition_variable>
+#include <cstring>
+#include <deque>
+#include <mutex>
+#include <stdexcept>
+#include <thread>
+#include <unordered_map>
+
+#include "core.h"
+#include "kernels.h"
+#include "timeline.h"
+
+namespace hvd {
+namespace gpu {
+
+namespace {
+
+#define HIP_CHECK(cmd)                                                      \
+  do {                                                                      \
+    hipError_t e_ = (cmd);                                                  \
+    if (e_ != hipSuccess)                                                   \
+      throw std::runtime_error(std::string("HIP error: ") +                 \
+                               hipGetErrorString(e_) + " at " #cmd);        \
+  } while (0)
+
+#define RCCL_CHECK(cmd)                                                     \
+  do {                                                                      \
+    ncclResult_t r_ = (cmd);                                                \
+    if (r_ != ncclSuccess)                                                  \
+      throw std::runtime_error(std::string("RCCL error: ") +                \
+                               ncclGetErrorString(r_) + " at " #cmd);       \
+  } while (0)
+
+ncclDataType_t ToNccl(DataType t) {
+  switch (t) {
+    case DataType::HVD_UINT8: return ncclUint8;
+    case DataType::HVD_INT8: return ncclInt8;
+    case DataType::HVD_INT32: return ncclInt32;
+    case DataType::HVD_INT64: return ncclInt64;
+    case DataType::HVD_FLOAT16: return ncclFloat16;
+    case DataType::HVD_FLOAT32: return ncclFloat32;
+    case DataType::HVD_FLOAT64: return ncclFloat64;
+    case DataType::HVD_BFLOAT16: return ncclBfloat16;
+    case DataType::HVD_BOOL: return ncclUint8;
+    default:
+      throw std::runtime_error("horovod_amd: dtype unsupported by RCCL");
+  }
+}
+
+ncclRedOp_t ToNcclOp(ReduceOp op) {
+  switch (op) {
+    case ReduceOp::SUM:
+    case ReduceOp::AVERAGE: return ncclSum;  // average via postscale
+    case ReduceOp::MIN: return ncclMin;
+    case ReduceOp::MAX: return ncclMax;
+    case ReduceOp::PRODUCT: return ncclProd;
+    default: throw std::runtime_error("horovod_amd: bad reduce op for RCCL");
+  }
+}
+
+struct DeviceCtx {
+  int device = -1;
+  c10::hip::HIPStream stream;  // dedicated comm stream (torch pool, high prio)
+  at::Tensor fusion_buffer;    // persistent, byte-typed
+  at::Tensor adasum_buffer;    // [set_size x fused] gather space for adasum
+  at::Tensor dots_buffer;      // adasum per-tensor {dot,|a|2,|b|2} doubles
+  std::unordered_map<int32_t, ncclComm_t> comms;  // process_set -> comm
+
+  explicit DeviceCtx(int dev)
+      : device(dev), stream(c10::hip::getStreamFromPool(true, dev)) {}
+};
+
+struct PendingOp {
+  hipEvent_t done_event = nullptr;
+  std::vector<TensorTableEntry> entries;
+  std::vector<hipEvent_t> ready_events;
+  int device = -1;
+  int64_t start_us = 0;
+  std::string activity;
+};
+
+struct Finalizer {
+  std::mutex mu;
+  std::condition_variable cv;
+  std::deque<PendingOp> queue;
+  std::thread thread;
+  bool stop = false;
+  bool started = false;
+} g_finalizer;
+
+std::mutex g_ctx_mu;
+std::unordered_map<int, DeviceCtx*> g_ctx;
+std::atomic<bool> g_rccl_used{false};
+
+void FinalizerLoop() {
+  while (true) {
+    PendingOp op;
+    {
+      std::unique_lock<std::mutex> lk(g_finalizer.mu);
+      g_finalizer.cv.wait(lk,
+                          [] { return g_finalizer.stop || !g_finalizer.queue.empty(); });
+      if (g_finalizer.queue.empty()) {
+        if (g_finalizer.stop) return;
+        continue;
+      }
+      op = std::move(g_finalizer.queue.front());
+      g_finalizer.queue.pop_front();
+    }
+    hipError_t e = hipEventSynchronize(op.done_event);
+    Status s = e == hipSuccess
+                   ? Status::OK()
+                   : Status::UnknownError(std::string("hipEventSynchronize: ") +
+                                          hipGetErrorString(e));
+    for (auto& entry : op.entries)
+      if (entry.callback) entry.callback(s, entry);
+    hipEventDestroy(op.done_event);
+    for (auto ev : op.ready_events) hipEventDestroy(ev);
+    auto& st = State();
+    if (st.timeline && !op.entries.empty())
+      st.timeline->Activity(op.entries[0].name, op.activity, op.start_us,
+                            st.timeline->NowUs());
+  }
+}
+
+void EnsureFinalizer() {
+  std::lock_guard<std::mutex> g(g_finalizer.mu);
+  if (!g_finalizer.started) {
+    g_finalizer.thread = std::thread(FinalizerLoop);
+    g_finalizer.started = true;
+  }
+}
+
+DeviceCtx& GetCtx(int device) {
+  std::lock_guard<std::mutex> g(g_ctx_mu);
+  auto it = g_ctx.find(device);
+  if (it != g_ctx.end()) return *it->second;
+  c10::hip::HIPGuard guard(device);
+  auto* ctx = new DeviceCtx(device);
+  g_ctx[device] = ctx;
+  return *ctx;
+}
+
+// uniqueId exchange over the TCP star — every global rank participates so
+// the lock-step framing holds even when rank 0 is not a set member.
+std::string ExchangeUniqueId(GlobalState& st, int leader,
+                             const std::string& payload) {
+  auto gathered = st.comm.Gather(payload);
+  std::string idb;
+  if (st.comm.is_root() && leader < (int)gathered.size()) idb = gathered[leader];
+  return st.comm.Bcast(idb);
+}
+
+// Lazy RCCL communicator for a process set (reference: nccl_operations.cc
+// 87-131 — id bcast via the controller, lock-step on every global rank).
+ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
+  auto it = ctx.comms.find(set_id);
+  if (it != ctx.comms.end()) return it->second;
+  auto& set = st.controller->process_set(set_id);
+  int leader = set.ranks.empty() ? 0 : set.ranks[0];
+
+  ncclUniqueId id;
+  std::string payload;
+  if (st.rank == leader) {
+    RCCL_CHECK(ncclGetUniqueId(&id));
+    payload.assign((const char*)&id, sizeof(id));
+  }
+  std::string idb = ExchangeUniqueId(st, leader, payload);
+  if (idb.size() != sizeof(id))
+    throw std::runtime_error("horovod_amd: RCCL uniqueId exchange failed");
+  std::memcpy(&id, idb.data(), sizeof(id));
+
+  c10::hip::HIPGuard guard(ctx.device);
+  ncclComm_t comm = nullptr;
+  RCCL_CHECK(ncclCommInitRank(&comm, (int)set.ranks.size(), id,
+                              set.local_index(st.rank)));
+  ctx.comms[set_id] = comm;
+  g_rccl_used = true;
+  return comm;
+}
+
+// Process sets whose RCCL comm bootstrap already ran on this rank (background
+// thread only — every rank passes the same first-response for a set).
+std::unordered_map<int32_t, bool> g_bootstrapped;
+
+at::Tensor& FusionBuffer(DeviceCtx& ctx, int64_t bytes) {
+  if (!ctx.fusion_buffer.defined() || ctx.fusion_buffer.numel() < bytes) {
+    c10::hip::HIPStreamGuard sg(ctx.stream);
+    ctx.fusion_buffer = at::empty(
+        {bytes}, at::TensorOptions().dtype(at::kByte).device(at::kCUDA, ctx.device));
+  }
+  return ctx.fusion_buffer;
+}
+
+void RecordStreamFor(const at::Tensor& t, const c10::hip::HIPStream& s) {
+  if (t.defined() && !t.is_cpu())
+    c10::hip::HIPCachingAllocator::recordStream(t.storage().data_ptr(), s);
+}
+
+void WaitReadyEvents(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
+                     std::vector<hipEvent_t>& ready) {
+  for (auto& e : entries) {
+    if (e.ready_event) {
+      HIP_CHECK(hipStreamWaitEvent(ctx.stream.stream(), (hipEvent_t)e.ready_event, 0));
+      ready.push_back((hipEvent_t)e.ready_event);
+      e.ready_event = 0;
+    }
+    RecordStreamFor(e.tensor, ctx.stream);
+    RecordStreamFor(e.output, ctx.stream);
+  }
+}
+
+void Finalize(DeviceCtx& ctx, std::vector<TensorTableEntry> entries,
+              std::vector<hipEvent_t> ready, const char* activity,
+              int64_t start_us) {
+  PendingOp op;
+  HIP_CHECK(hipEventCreateWithFlags(&op.done_event, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(op.done_event, ctx.stream.stream()));
+  op.entries = std::move(entries);
+  op.ready_events = std::move(ready);
+  op.device = ctx.device;
+  op.activity = activity;
+  op.start_us = start_us;
+  EnsureFinalizer();
+  {
+    std::lock_guard<std::mutex> g(g_finalizer.mu);
+    g_finalizer.queue.push_back(std::move(op));
+  }
+  g_finalizer.cv.notify_one();
+}
+
+// Pack entries (with prescale + dtype conversion) into the fusion buffer at
+// 64-element-aligned offsets; returns total wire elements.
+int64_t PackEntries(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
+                    DataType wire, bool unpack, char* base_override = nullptr) {
+  int64_t wire_size = (int64_t)DataTypeSize(wire);
+  int64_t total = 0;
+  for (auto& e : entries) total += AlignedElems(e.tensor.numel());
+  char* base = base_override;
+  if (!base) {
+    auto& buf = FusionBuffer(ctx, total * wire_size);
+    base = (char*)buf.data_ptr();
+  }
+
+  CopyBatchArgs args;
+  bool any_scale = false;
+  int64_t off = 0;
+  int src_dt = 0, dst_dt = 0;
+  auto flush = [&](int s_dt, int d_dt) {
+    if (args.count == 0) return;
+    HIP_CHECK(BatchedCopyLaunch(args, s_dt, d_dt, any_scale, 32, ctx.stream.stream()));
+    args.count = 0;
+    any_scale = false;
+  };
+  for (auto& e : entries) {
+    at::Tensor t = unpack ? e.output : e.tensor;
+    if (!t.defined()) t = e.tensor;
+    int this_src = unpack ? (int)wire : (int)DataTypeFromTorch(t.scalar_type());
+    int this_dst = unpack ? (int)DataTypeFromTorch(t.scalar_type()) : (int)wire;
+    if (args.count == kCopyBatchCapacity ||
+        (args.count > 0 && (this_src != src_dt || this_dst != dst_dt)))
+      flush(src_dt, dst_dt);
+    src_dt = this_src;
+    dst_dt = this_dst;
+    double scale = unpack ? e.postscale : e.prescale;
+    int i = args.count++;
+    if (unpack) {
+      args.src[i] = base + off * wire_size;
+      args.dst[i] = t.data_ptr();
+    } else {
+      args.src[i] = t.data_ptr();
+      args.dst[i] = base + off * wire_size;
+    }
+    args.numel[i] = (unsigned long long)t.numel();
+    args.scale[i] = scale;
+    if (scale != 1.0) any_scale = true;
+    off += AlignedElems(t.numel());
+  }
+  flush(src_dt, dst_dt);
+  return total;
+}
+
+
+// One-shot Adasum over xGMI: allgather every rank's fused buffer (7 wide
+// point-to-point links make this efficient on one MI355X node), then run the
+// VHDD combine tree LOCALLY with the CDNA4 dot/scaled-add kernels — a
+// bitwise-identical result on every rank with zero host synchronization.
+// (Reference: AdasumGpuAllreduceOp, adasum_gpu_operations.cc:44-120, which
+// does NCCL RS + host-MPI VHDD + NCCL AG; ours keeps everything on-device.)
+void ExecuteAdasum(GlobalState& st, DeviceCtx& ctx, Response& resp,
+                   std::vector<TensorTableEntry>& entries, ncclComm_t comm) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  int n = (int)set.ranks.size();
+  DataType wire = resp.dtype;
+  int64_t wire_size = (int64_t)DataTypeSize(wire);
+  hipStream_t stream = ctx.stream.stream();
+
+  int64_t total = PackEntries(ctx, entries, wire, false);
+  char* src_base = (char*)ctx.fusion_buffer.data_ptr();
+  char* result_base = src_base;
+  if (n > 1) {
+    int64_t row_bytes = total * wire_size;
+    if (!ctx.adasum_buffer.defined() ||
+        ctx.adasum_buffer.numel() < (int64_t)n * row_bytes) {
+      c10::hip::HIPStreamGuard sg(ctx.stream);
+      ctx.adasum_buffer =
+          at::empty({(int64_t)n * row_bytes},
+                    at::TensorOptions().dtype(at::kByte).device(at::kCUDA, ctx.device));
+    }
+    char* ab = (char*)ctx.adasum_buffer.data_ptr();
+    RCCL_CHECK(ncclAllGather(src_base, ab, total, ToNccl(wire), comm, stream));
+
+    int64_t ndots = (int64_t)entries.size() * 3;
+    if (!ctx.dots_buffer.defined() || ctx.dots_buffer.numel() < ndots) {
+      c10::hip::HIPStreamGuard sg(ctx.stream);
+      ctx.dots_buffer = at::empty(
+          {ndots}, at::TensorOptions().dtype(at::kDouble).device(at::kCUDA, ctx.device));
+    }
+    double* dots = ctx.dots_buffer.data_ptr<double>();
+
+    // per-tensor offsets within a row (64-elem aligned, matching PackEntries)
+    std::vector<int64_t> offs, cnts;
+    int64_t off = 0;
+    for (auto& e : entries) {
+      offs.push_back(off);
+      cnts.push_back(e.tensor.numel());
+      off += AlignedElems(e.tensor.numel());
+    }
+    auto combine = [&](int i, int j) {
+      char* rowa = ab + (int64_t)i * row_bytes;
+      char* rowb = ab + (int64_t)j * row_bytes;
+      HIP_CHECK(hipMemsetAsync(dots, 0, (size_t)ndots * sizeof(double), stream));
+      for (size_t start = 0; start < entries.size(); start += kCopyBatchCapacity) {
+        AdasumBatchArgs a;
+        a.count = (int)std::min<size_t>(kCopyBatchCapacity, entries.size() - start);
+        for (int k = 0; k < a.count; ++k) {
+          a.a[k] = rowa + offs[start + k] * wire_size;
+          a.b[k] = rowb + offs[start + k] * wire_size;
+          a.numel[k] = (unsigned long long)cnts[start + k];
+        }
+        HIP_CHECK(AdasumDotsLaunch(a, (int)wire, dots + start * 3, stream));
+      }
+      for (size_t start = 0; start < entries.size(); start += kCopyBatchCapacity) {
+        AdasumBatchArgs a;
+        a.count = (int)std::min<size_t>(kCopyBatchCapacity, entries.size() - start);
+        for (int k = 0; k < a.count; ++k) {
+          a.a[k] = rowa + offs[start + k] * wire_size;
+          a.b[k] = rowb + offs[start + k] * wire_size;
+          a.numel[k] = (unsigned long long)cnts[start + k];
+        }
+        HIP_CHECK(AdasumScaledAddLaunch(a, (int)wire, dots + start * 3, stream));
+      }
+    };
+    // VHDD tree (matches the CPU/golden ordering in core.cc): fold the
+    // non-power-of-2 remainder, then pairwise distance-doubling.
+    int p = 1;
+    while (p * 2 <= n) p *= 2;
+    for (int i = p; i < n; ++i) combine(i - p, i);
+    for (int stride = 1; stride < p; stride *= 2)
+      for (int i = 0; i + stride < p; i += 2 * stride) combine(i, i + stride);
+    result_base = ab;  // row 0
+  }
+  for (auto& e : entries)
+    if (!e.output.defined()) {
+      c10::hip::HIPStreamGuard sg(ctx.stream);
+      e.output = at::empty_like(e.tensor);
+    }
+  PackEntries(ctx, entries, wire, true, result_base);
+}
+
+}  // namespace
+
+uintptr_t RecordReadyEvent(int device) {
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, c10::hip::getCurrentHIPStream(device).stream()));
+  return (uintptr_t)ev;
+}
+
+bool RcclUsed() { return g_rccl_used; }
+
+void Execute(GlobalState& st, Response& resp,
+             std::vector<TensorTableEntry>& entries) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  if (!g_bootstrapped.count(resp.process_set_id)) {
+    // Every global rank reaches this point on the set's first GPU response
+    // (responses are broadcast in a fixed order), so members can init the
+    // RCCL comm while non-members relay the uniqueId frames in lock-step.
+    if (!entries.empty()) {
+      EnsureComm(st, GetCtx(entries[0].device), resp.process_set_id);
+    } else {
+      ExchangeUniqueId(st, set.ranks.empty() ? 0 : set.ranks[0], "");
+    }
+    g_bootstrapped[resp.process_set_id] = true;
+  }
+  if (entries.empty()) return;  // relay-only rank (not a member)
+  int device = entries[0].device;
+  auto& ctx = GetCtx(device);
+  ncclComm_t comm = EnsureComm(st, ctx, resp.process_set_id);
+  c10::hip::HIPGuard dguard(device);
+  hipStream_t stream = ctx.stream.stream();
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  DataType wire = resp.dtype;
+  int64_t wire_size = (int64_t)DataTypeSize(wire);
+  auto wire_nccl = ToNccl(wire);
+  int64_t t_start = st.timeline ? st.timeline->NowUs() : 0;
+
+  std::vector<hipEvent_t> ready;
+  WaitReadyEvents(ctx, entries, ready);
+
+  const char* activity = "RCCL_OP";
+  switch (resp.type) {
+    case ResponseType::ALLREDUCE: {
+      activity = "RCCL_ALLREDUCE";
+      bool direct = entries.size() == 1 && entries[0].prescale == 1.0 &&
+                    entries[0].postscale == 1.0 &&
+                    DataTypeFromTorch(entries[0].tensor.scalar_type()) == wire &&
+                    entries[0].tensor.is_contiguous() &&
+                    entries[0].output.defined() && entries[0].output.is_contiguous();
+      if (direct) {
+        auto& e = entries[0];
+        RCCL_CHECK(ncclAllReduce(e.tensor.data_ptr(), e.output.data_ptr(),
+                                 e.tensor.numel(), wire_nccl,
+                                 ToNcclOp(resp.reduce_op), comm, stream));
+      } else {
+        int64_t total = PackEntries(ctx, entries, wire, false);
+        RCCL_CHECK(ncclAllReduce(ctx.fusion_buffer.data_ptr(),
+                                 ctx.fusion_buffer.data_ptr(), total, wire_nccl,
+                                 ToNcclOp(resp.reduce_op), comm, stream));
+        for (auto& e : entries)
+          if (!e.output.defined()) {
+            c10::hip::HIPStreamGuard sg(ctx.stream);
+            e.output = at::empty_like(e.tensor);
+          }
+        PackEntries(ctx, entries, wire, true);
+      }
+      break;
+    }
+    case ResponseType::ADASUM: {
+      activity = "ADASUM";
+      ExecuteAdasum(st, ctx, resp, entries, comm);
+      break;
+    }
+    case ResponseType::BROADCAST: {
+      activity = "RCCL_BCAST";
+      auto& e = entries[0];
+      int root_li = set.local_index(resp.root_rank);
+      at::Tensor in = e.tensor.is_contiguous() ? e.tensor : e.tensor.contiguous();
+      if (!e.output.defined()) e.output = e.tensor;
+      RCCL_CHECK(ncclBroadcast(in.data_ptr(), e.output.data_ptr(), in.numel(),
+                               wire_nccl, root_li, comm, stream));
+      break;
+    }
+    case ResponseType::ALLGATHER: {
+      activity = "RCCL_ALLGATHER";
+      auto& e = entries[0];
+      at::Tensor in = e.tensor.contiguous();
+      int64_t row_elems = e.tensor.numel();
+      if (e.tensor.dim() > 0 && e.tensor.size(0) > 0)
+        row_elems = e.tensor.numel() / e.tensor.size(0);
+      int64_t total0 = 0;
+      bool same = true;
+      for (int r = 0; r < n; ++r) {
+        total0 += resp.tensor_sizes[r];
+        if (resp.tensor_sizes[r] != resp.tensor_sizes[0]) same = false;
+      }
+      std::vector<int64_t> out_shape(e.tensor.sizes().begin(),
+                                     e.tensor.sizes().end());
+      if (out_shape.empty()) out_shape = {total0};
+      else out_shape[0] = total0;
+      {
+        c10::hip::HIPStreamGuard sg(ctx.stream);
+        e.output = at::empty(out_shape, e.tensor.options());
+      }
+      if (same && total0 > 0) {
+        RCCL_CHECK(ncclAllGather(in.data_ptr(), e.output.data_ptr(),
+                                 resp.tensor_sizes[0] * row_elems, wire_nccl, comm,
+                                 stream));
+      } else {
+        // v-variant via grouped broadcasts (reference: nccl_operations.cc
+        // 1082-1101)
+        RCCL_CHECK(ncclGroupStart());
+        char* out_base = (char*)e.output.data_ptr();
+        int64_t off = 0;
+        for (int r = 0; r < n; ++r) {
+          int64_t cnt = resp.tensor_sizes[r] * row_elems;
+          if (cnt > 0) {
+            RCCL_CHECK(ncclBroadcast(
+                r == li ? in.data_ptr() : (void*)(out_base + off * wire_size),
+                out_base + off * wire_size, cnt, wire_nccl, r, comm, stream));
+          }
+          off += cnt;
+        }
+        RCCL_CHECK(ncclGroupEnd());
+      }
+      break;
+    }
+    case ResponseType::ALLTOALL: {
+      activity = "RCCL_ALLTOALL";
+      auto& e = entries[0];
+      at::Tensor in = e.tensor.contiguous();
+      int64_t row_elems = 1;
+      for (int d = 1; d < e.tensor.dim(); ++d) row_elems *= e.tensor.size(d);
+      // splits matrix: row i = sender i's splits
+      int64_t recv_rows = 0;
+      for (int r = 0; r < n; ++r) recv_rows += resp.tensor_sizes[(size_t)r * n + li];
+      std::vector<int64_t> out_shape(e.tensor.sizes().begin(),
+                                     e.tensor.sizes().end());
+      if (out_shape.empty()) out_shape = {recv_rows};
+      else out_shape[0] = recv_rows;
+      {
+        c10::hip::HIPStreamGuard sg(ctx.stream);
+        e.output = at::empty(out_shape, e.tensor.options());
+      }
+      char* in_base = (char*)in.data_ptr();
+      char* out_base = (char*)e.output.data_ptr();
+      RCCL_CHECK(ncclGroupStart());
+      int64_t send_off = 0, recv_off = 0;
+      for (int r = 0; r < n; ++r) {
+        int64_t s_rows = resp.tensor_sizes[(size_t)li * n + r];
+        int64_t r_rows = resp.tensor_sizes[(size_t)r * n + li];
+        if (s_rows > 0)
+          RCCL_CHECK(ncclSend(in_base + send_off * row_elems * wire_size,
+                              s_rows * row_elems, wire_nccl, r, comm, stream));
+        if (r_rows > 0)
+          RCCL_CHECK(ncclRecv(out_base + recv_off * row_elems * wire_size,
+                              r_rows * row_elems, wire_nccl, r, comm, stream));
+        send_off += s_rows;
+        recv_off += r_rows;
+      }
+      RCCL_CHECK(ncclGroupEnd());
+      e.received_splits = at::empty({n}, at::kLong);
+      auto* rs = e.received_splits.data_ptr<int64_t>();
+      for (int r = 0; r < n; ++r) rs[r] = resp.tensor_sizes[(size_t)r * n + li];
+      break;
+    }
+    case ResponseType::REDUCESCATTER: {
+      activity = "RCCL_REDUCESCATTER";
+      auto& e = entries[0];
+      at::Tensor in = e.tensor.contiguous();
+      int64_t first = e.tensor.dim() > 0 ? e.tensor.size(0) : 1;
+      int64_t row_elems = first > 0 ? e.tensor.numel() / first : 0;
+      int64_t base_rows = first / n, rem = first % n;
+      int64_t my_rows = base_rows + (li < rem ? 1 : 0);
+      std::vector<int64_t> out_shape(e.tensor.sizes().begin(),
+                                     e.tensor.sizes().end());
+      if (out_shape.empty()) out_shape = {my_rows};
+      else out_shape[0] = my_rows;
+      {
+        c10::hip::HIPStreamGuard sg(ctx.stream);
+        e.output = at::empty(out_shape, e.tensor.options());
+      }
+      if (rem == 0 && first > 0) {
+        RCCL_CHECK(ncclReduceScatter(in.data_ptr(), e.output.data_ptr(),
+                                     base_rows * row_elems, wire_nccl,
+                                     ToNcclOp(resp.reduce_op), comm, stream));
+      } else {
+        // v-variant via grouped reduces (reference: nccl_operations.cc
+        // 1294-1316)
+        char* in_base = (char*)in.data_ptr();
+        RCCL_CHECK(ncclGroupStart());
+        int64_t off = 0;
+        for (int r = 0; r < n; ++r) {
+          int64_t rows = base_rows + (r < rem ? 1 : 0);
+          if (rows > 0)
+            RCCL_CHECK(ncclReduce(in_base + off * row_elems * wire_size,
+                                  e.output.data_ptr(), rows * row_elems, wire_nccl,
+                                  ToNcclOp(resp.reduce_op), r, comm, stream));
+          off += rows;
+        }
+        RCCL_CHECK(ncclGroupEnd());
+      }
+      if (e.postscale != 1.0) {
+        CopyBatchArgs args;
+        args.count = 1;
+        args.src[0] = e.output.data_ptr();
+        args.dst[0] = e.output.data_ptr();
+        args.numel[0] = (unsigned long long)e.output.numel();
+        args.scale[0] = e.postscale;
+        HIP_CHECK(BatchedCopyLaunch(args, (int)wire, (int)wire, true, 32, stream));
+      }
+      break;
+    }
+    default:
+      throw std::runtime_error("horovod_amd: bad GPU response type");
+  }
+
+  Finalize(ctx, std::move(entries), std::move(ready), activity, t_start);
+}
+
+void WaitAllPending() {
+  while (true) {
+    {
+      std::lock_guard<std::mutex> g(g_finalizer.mu);
+      if (g_finalizer.queue.empty()) break;
+    }
+    std::this_thread::sleep_for(std::chrono::milliseconds(1));
+  }
+}
+
+void Shutdown() {
+  WaitAllPending();
+  {
+    std::lock_guard<std::mutex> g(g_finalizer.mu);
+    g_finalizer.stop = true;
+  }
+  g_finalizer.cv.notify_all();
+  if (g_finalizer.started && g_finalizer.thread.joinable()) g_finalizer.thread.join();
+  g_finalizer.started = false;
+  g_finalizer.stop = false;
+  std::lock_guard<std::mutex> g(g_ctx_mu);
+  for (auto& kv : g_ctx) {
+    for (auto& ck : kv.second->comms) ncclCommDestroy(ck.second);
+    delete kv.second;
+  }
+  g_ctx.clear();
+  g_bootstrapped.clear();
+}
+
+}  // namespace gpu
+}  // namespace hvd

@@ -1,0 +1,39 @@
+// GPU data plane: RCCL over xGMI on a dedicated high-priority HIP stream.
+//
+// Re-design of the reference's GPUContext/GPUOpContext + NCCL op set
+// (horovod/common/ops/gpu_operations.{cc,h} N7, nccl_operations.{cc,h} N8)
+// for MI355X: one process per GPU, RCCL as the only backend, torch's HIP
+// stream pool for allocator-safe stream interop, pack/unpack through the
+// CDNA4 batched-copy kernels (kernels.hip), async completion via a finalizer
+// thread.
+#pragma once
+
+#include <cstdint>
+#include <vector>
+
+namespace hvd {
+struct GlobalState;
+struct Response;
+struct TensorTableEntry;
+
+namespace gpu {
+
+// Record an event on torch's current stream for `device` (the producer
+// ordering point the comm stream will wait on).  Returns an opaque handle.
+uintptr_t RecordReadyEvent(int device);
+
+// Execute a (possibly fused) response on the GPU.  Asynchronous: entries'
+// callbacks fire from the finalizer thread once the comm-stream work is done.
+void Execute(GlobalState& st, Response& resp,
+             std::vector<TensorTableEntry>& entries);
+
+// Drain the finalizer queue (shutdown path).
+void WaitAllPending();
+void Shutdown();
+
+// True once a RCCL communicator has been created (used by tests to assert
+// the native path ran).
+bool RcclUsed();
+
+}  // namespace gpu
+}  // namespace hvd

@@ -1,0 +1,61 @@
+// Host-visible interface to the CDNA4 kernels (kernels.hip).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+namespace hvd {
+namespace gpu {
+
+// Matches hvd::DataType codes (common.h) — kept as plain ints so this header
+// stays torch-free for the .hip TU.
+enum KDType {
+  DT_U8 = 0,
+  DT_I8 = 1,
+  DT_I32 = 2,
+  DT_I64 = 3,
+  DT_F16 = 4,
+  DT_F32 = 5,
+  DT_F64 = 6,
+  DT_BOOL = 7,
+  DT_BF16 = 8,
+  DT_U16 = 9,
+  DT_I16 = 10,
+};
+
+// Descriptor capacity per launch: sized so the by-value kernarg stays well
+// under the 4 KB limit (48 * 28 B + 8 ≈ 1.4 KB).
+constexpr int kCopyBatchCapacity = 48;
+
+struct CopyBatchArgs {
+  const void* src[kCopyBatchCapacity];
+  void* dst[kCopyBatchCapacity];
+  unsigned long long numel[kCopyBatchCapacity];
+  double scale[kCopyBatchCapacity];
+  int count = 0;
+};
+
+// Batched strided copy with optional per-entry scale and src->dst dtype
+// conversion.  grid = count * blocks_per_copy blocks of 256 threads.
+hipError_t BatchedCopyLaunch(const CopyBatchArgs& args, int src_dt, int dst_dt,
+                             bool with_scale, int blocks_per_copy,
+                             hipStream_t stream);
+
+// ---- Adasum device pipeline (adasum_kernels.hip) ---------------------------
+// Stage 1: per-tensor double-precision dot products / squared norms of (a,b)
+// pairs; Stage 2: a = acoef*a + bcoef*b with coefficients derived on-device.
+struct AdasumBatchArgs {
+  void* a[kCopyBatchCapacity];        // dtype = wire dtype
+  const void* b[kCopyBatchCapacity];
+  unsigned long long numel[kCopyBatchCapacity];
+  int count = 0;
+};
+
+// dots layout: [count][3] doubles = {dot(a,b), |a|^2, |b|^2}; must be zeroed
+// before the dot pass.
+hipError_t AdasumDotsLaunch(const AdasumBatchArgs& args, int dt, double* dots,
+                            hipStream_t stream);
+hipError_t AdasumScaledAddLaunch(const AdasumBatchArgs& args, int dt,
+                                 const double* dots, hipStream_t stream);
+
+}  // namespace gpu
+}  // namespace hvd
