@@ -1,0 +1,225 @@
+"""CPU multi-process collective tests (reference coverage model:
+test/parallel/test_torch.py — every op x dtype x variant under a real
+multi-process launch)."""
+import pytest
+import torch
+
+from tests.parallel_util import run_workers
+
+
+def test_allreduce_sum_np2():
+    run_workers(2, """
+        t = torch.arange(10, dtype=torch.float32) * (rank + 1)
+        out = hvd.allreduce(t, average=False)
+        expected = torch.arange(10, dtype=torch.float32) * 3
+        assert torch.allclose(out, expected), (out, expected)
+    """)
+
+
+def test_allreduce_average_np3():
+    run_workers(3, """
+        t = torch.ones(7) * (rank + 1)
+        out = hvd.allreduce(t)  # default Average
+        assert torch.allclose(out, torch.full((7,), 2.0)), out
+    """)
+
+
+def test_allreduce_dtypes_np2():
+    run_workers(2, """
+        for dtype in [torch.float32, torch.float64, torch.int32, torch.int64,
+                      torch.uint8, torch.int8, torch.float16, torch.bfloat16]:
+            t = torch.arange(5).to(dtype)
+            out = hvd.allreduce(t, average=False,
+                                name=f"dt.{str(dtype)}")
+            assert torch.allclose(out.float(), torch.arange(5).float() * 2), \
+                (dtype, out)
+    """)
+
+
+def test_allreduce_min_max_product_np2():
+    run_workers(2, """
+        t = torch.tensor([1.0, 5.0]) if rank == 0 else torch.tensor([3.0, 2.0])
+        mn = hvd.allreduce(t, op=hvd.Min, name="mn")
+        mx = hvd.allreduce(t, op=hvd.Max, name="mx")
+        pr = hvd.allreduce(t, op=hvd.Product, name="pr")
+        assert torch.allclose(mn, torch.tensor([1.0, 2.0])), mn
+        assert torch.allclose(mx, torch.tensor([3.0, 5.0])), mx
+        assert torch.allclose(pr, torch.tensor([3.0, 10.0])), pr
+    """)
+
+
+def test_allreduce_prescale_postscale_np2():
+    run_workers(2, """
+        t = torch.ones(4)
+        out = hvd.allreduce(t, average=False, prescale_factor=2.0,
+                            postscale_factor=0.5, name="scaled")
+        # (1*2 + 1*2) * 0.5 = 2
+        assert torch.allclose(out, torch.full((4,), 2.0)), out
+    """)
+
+
+def test_allreduce_inplace_and_async_np2():
+    run_workers(2, """
+        t = torch.ones(3)
+        hvd.allreduce_(t, average=False, name="inplace")
+        assert torch.allclose(t, torch.full((3,), 2.0)), t
+        h = hvd.allreduce_async(torch.ones(3), average=False, name="async1")
+        assert hvd.synchronize(h).sum().item() == 6.0
+        h2 = hvd.allreduce_async_(torch.ones(3), average=False, name="async2")
+        while not hvd.poll(h2):
+            pass
+        assert hvd.synchronize(h2).sum().item() == 6.0
+    """)
+
+
+def test_grouped_allreduce_np2():
+    run_workers(2, """
+        ts = [torch.ones(5) * (rank + 1), torch.ones(3) * (rank + 10)]
+        outs = hvd.grouped_allreduce(ts, average=False, name="grp")
+        assert torch.allclose(outs[0], torch.full((5,), 3.0)), outs[0]
+        assert torch.allclose(outs[1], torch.full((3,), 21.0)), outs[1]
+    """)
+
+
+def test_allgather_same_shape_np2():
+    run_workers(2, """
+        t = torch.arange(6).reshape(3, 2).float() * (rank + 1)
+        out = hvd.allgather(t)
+        assert out.shape == (6, 2), out.shape
+        assert torch.allclose(out[:3], torch.arange(6).reshape(3, 2).float())
+        assert torch.allclose(out[3:], torch.arange(6).reshape(3, 2).float()*2)
+    """)
+
+
+def test_allgather_variable_shape_np3():
+    run_workers(3, """
+        n = rank + 1
+        t = torch.full((n, 2), float(rank))
+        out = hvd.allgather(t, name="agv")
+        assert out.shape == (6, 2), out.shape
+        assert torch.allclose(out[0:1], torch.zeros(1, 2))
+        assert torch.allclose(out[1:3], torch.ones(2, 2))
+        assert torch.allclose(out[3:6], torch.full((3, 2), 2.0))
+    """)
+
+
+def test_broadcast_np2():
+    run_workers(2, """
+        t = torch.arange(5).float() * (rank + 1)
+        out = hvd.broadcast(t, root_rank=1, name="bc")
+        assert torch.allclose(out, torch.arange(5).float() * 2), out
+        t2 = torch.full((3,), float(rank))
+        hvd.broadcast_(t2, root_rank=0, name="bc2")
+        assert torch.allclose(t2, torch.zeros(3)), t2
+    """)
+
+
+def test_alltoall_np2():
+    run_workers(2, """
+        # rank r sends rows [r*2, r*2+1) to each peer
+        t = torch.arange(8).reshape(4, 2).float() + rank * 100
+        out, rsplits = hvd.alltoall(t, splits=[2, 2], name="a2a")
+        assert out.shape == (4, 2), out.shape
+        if rank == 0:
+            expected = torch.cat([torch.arange(4).reshape(2, 2).float(),
+                                  torch.arange(4).reshape(2, 2).float() + 100])
+        else:
+            expected = torch.cat([torch.arange(4, 8).reshape(2, 2).float(),
+                                  torch.arange(4, 8).reshape(2,2).float()+100])
+        assert torch.allclose(out, expected), (out, expected)
+        assert rsplits.tolist() == [2, 2]
+    """)
+
+
+def test_alltoall_uneven_np2():
+    run_workers(2, """
+        if rank == 0:
+            t = torch.arange(3).float()
+            splits = [1, 2]
+        else:
+            t = torch.arange(10, 13).float()
+            splits = [2, 1]
+        out, rsplits = hvd.alltoall(t, splits=splits, name="a2av")
+        if rank == 0:
+            assert out.tolist() == [0.0, 10.0, 11.0], out
+            assert rsplits.tolist() == [1, 2]
+        else:
+            assert out.tolist() == [1.0, 2.0, 12.0], out
+            assert rsplits.tolist() == [2, 1]
+    """)
+
+
+def test_reducescatter_np2():
+    run_workers(2, """
+        t = torch.arange(8).float().reshape(4, 2) * (rank + 1)
+        out = hvd.reducescatter(t, op=hvd.Sum, name="rs")
+        full = torch.arange(8).float().reshape(4, 2) * 3
+        mine = full[rank * 2:(rank + 1) * 2]
+        assert torch.allclose(out, mine), (out, mine)
+    """)
+
+
+def test_reducescatter_uneven_np2():
+    run_workers(2, """
+        t = torch.arange(6).float() * (rank + 1)  # first dim 6... wait 1-D len 6? use 5
+        t = torch.arange(5).float() * (rank + 1)
+        out = hvd.reducescatter(t, op=hvd.Sum, name="rsv")
+        full = torch.arange(5).float() * 3
+        if rank == 0:
+            assert torch.allclose(out, full[:3]), out
+        else:
+            assert torch.allclose(out, full[3:]), out
+    """)
+
+
+def test_barrier_np3():
+    run_workers(3, """
+        import time
+        time.sleep(0.1 * rank)
+        hvd.barrier()
+    """)
+
+
+def test_join_np2():
+    run_workers(2, """
+        # rank 0 runs 3 batches, rank 1 runs 1 batch; join pads the gap
+        nb = 3 if rank == 0 else 1
+        for i in range(nb):
+            out = hvd.allreduce(torch.ones(4), average=False, name=f"b{i}")
+        last = hvd.join()
+        assert last == 0 or last == 1
+    """, timeout=240)
+
+
+def test_process_sets_np3():
+    run_workers(3, """
+        ps = hvd.add_process_set(hvd.ProcessSet([0, 2]))
+        if rank in (0, 2):
+            out = hvd.allreduce(torch.ones(3) * (rank + 1), average=False,
+                                process_set=ps, name="sub")
+            assert torch.allclose(out, torch.full((3,), 4.0)), out
+        # global op still works for everyone
+        out = hvd.allreduce(torch.ones(2), average=False, name="glob")
+        assert out.sum().item() == 6.0
+    """)
+
+
+def test_error_mismatched_shape_np2():
+    run_workers(2, """
+        n = 4 if rank == 0 else 5
+        try:
+            hvd.allreduce(torch.ones(n), average=False, name="bad")
+            raise SystemExit("expected an error for mismatched shapes")
+        except RuntimeError as e:
+            assert "Mismatched" in str(e), e
+    """)
+
+
+def test_steady_state_cache_np2():
+    """Many iterations of the same tensors: exercises the response-cache
+    fast path (steady-state training must not touch the coordinator)."""
+    run_workers(2, """
+        for i in range(50):
+            out = hvd.allreduce(torch.ones(1000), average=False, name="steady")
+            assert out.sum().item() == 2000.0
+    """)
