@@ -1,0 +1,143 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 synthetic-ImageNet training throughput.
+
+Matches BASELINE.json's metric ("images/sec/GPU + scaling efficiency,
+ResNet-50 synthetic at 1/2/4/8 MI355X") and the reference's benchmark method
+(examples/pytorch/pytorch_synthetic_benchmark.py: synthetic data, batch
+64/GPU, data-parallel hvd.DistributedOptimizer).
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+One JSON line on rank 0; `value` is the WHOLE-JOB total images/sec.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=64, help="per-GPU batch")
+    p.add_argument("--model", default="resnet50")
+    p.add_argument("--image-size", type=int, default=224)
+    p.add_argument("--compression", default="none",
+                   choices=["none", "fp16", "bf16"])
+    p.add_argument("--use-adasum", action="store_true")
+    p.add_argument("--no-bf16", action="store_true",
+                   help="disable bf16 autocast (fp32 compute)")
+    p.add_argument("--no-overlap", action="store_true",
+                   help="disable backward/allreduce overlap (ablation)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import horovod_amd.torch as hvd
+    hvd.init()
+
+    cuda = torch.cuda.is_available()
+    if cuda:
+        torch.cuda.set_device(hvd.local_rank())
+        torch.backends.cudnn.benchmark = True
+        device = torch.device("cuda", hvd.local_rank())
+    else:
+        device = torch.device("cpu")
+
+    from horovod_amd.models import resnet50, resnet101, resnet152
+    model_fn = {"resnet50": resnet50, "resnet101": resnet101,
+                "resnet152": resnet152}[args.model]
+    torch.manual_seed(42)
+    model = model_fn().to(device)
+    if cuda:
+        model = model.to(memory_format=torch.channels_last)
+
+    batch = args.batch_size if cuda else 8
+    image_size = args.image_size if cuda else 64
+    data = torch.randn(batch, 3, image_size, image_size, device=device)
+    if cuda:
+        data = data.to(memory_format=torch.channels_last)
+    target = torch.randint(0, 1000, (batch,), device=device)
+
+    from horovod_amd.torch.compression import Compression
+    compression = {"none": Compression.none, "fp16": Compression.fp16,
+                   "bf16": Compression.bf16}[args.compression]
+
+    opt = torch.optim.SGD(model.parameters(), lr=0.0125 * hvd.size(),
+                          momentum=0.9, weight_decay=5e-5)
+    opt = hvd.DistributedOptimizer(
+        opt, named_parameters=model.named_parameters(),
+        compression=compression,
+        op=hvd.Adasum if args.use_adasum else hvd.Average)
+    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+    hvd.broadcast_optimizer_state(opt, root_rank=0)
+
+    loss_fn = torch.nn.CrossEntropyLoss()
+    use_bf16 = cuda and not args.no_bf16
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+            out = model(data)
+            loss = loss_fn(out, target)
+        loss.backward()
+        opt.step()
+
+    for _ in range(args.warmup):
+        step()
+
+    hvd.barrier()
+    if cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if cuda:
+        torch.cuda.synchronize()
+    hvd.barrier()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    elapsed_t = torch.tensor([elapsed], dtype=torch.float64)
+    elapsed = float(hvd.allreduce(elapsed_t, op=hvd.Max,
+                                  name="bench_elapsed").item())
+
+    n = hvd.size()
+    total_images = n * batch * args.steps
+    value = total_images / elapsed
+    if hvd.rank() == 0:
+        result = {
+            "metric": "images/sec, ResNet-50 synthetic (total over all GPUs)",
+            "value": round(value, 2),
+            "unit": "images/sec",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_bf16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": n * batch,
+                "image_size": image_size,
+                "parallelism": f"dp{n}",
+                "compression": args.compression,
+                "reduction": "adasum" if args.use_adasum else "average",
+            },
+        }
+        print(json.dumps(result), flush=True)
+    hvd.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,3 @@
+from horovod_amd.models.resnet import (ResNet, resnet50, resnet101,  # noqa: F401
+                                       resnet152)
+from horovod_amd.models.mlp import MNISTNet  # noqa: F401

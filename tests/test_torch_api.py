@@ -1,0 +1,243 @@
+"""Framework-level tests: DistributedOptimizer training convergence across
+ranks, state broadcast helpers, SyncBatchNorm numerics, compression,
+autograd of collectives, Adasum golden numerics.
+(Reference coverage model: test/parallel/test_torch.py + test_adasum_pytorch.py.)
+"""
+import numpy as np
+import pytest
+import torch
+
+from tests.parallel_util import run_workers
+
+
+def test_distributed_optimizer_sync_np2():
+    """After N steps on different data, parameters must be identical across
+    ranks and follow the average-gradient trajectory."""
+    run_workers(2, """
+        torch.manual_seed(1234)           # same init on all ranks
+        model = torch.nn.Sequential(
+            torch.nn.Linear(10, 16), torch.nn.ReLU(), torch.nn.Linear(16, 1))
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        opt = hvd.DistributedOptimizer(
+            opt, named_parameters=model.named_parameters())
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(100 + rank)     # different data per rank
+        for step in range(5):
+            x = torch.randn(8, 10)
+            y = torch.randn(8, 1)
+            opt.zero_grad()
+            loss = ((model(x) - y) ** 2).mean()
+            loss.backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        gathered = hvd.allgather(flat.unsqueeze(0), name="final_params")
+        assert torch.allclose(gathered[0], gathered[1], atol=1e-6), \
+            (gathered[0] - gathered[1]).abs().max()
+    """)
+
+
+def test_distributed_optimizer_groups_np2():
+    run_workers(2, """
+        torch.manual_seed(7)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(6, 8), torch.nn.ReLU(), torch.nn.Linear(8, 2))
+        opt = hvd.DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1),
+            named_parameters=model.named_parameters(), groups=2)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(100 + rank)
+        for step in range(3):
+            opt.zero_grad()
+            loss = model(torch.randn(4, 6)).sum()
+            loss.backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        g = hvd.allgather(flat.unsqueeze(0), name="gp")
+        assert torch.allclose(g[0], g[1], atol=1e-6)
+    """)
+
+
+def test_backward_passes_per_step_np2():
+    run_workers(2, """
+        torch.manual_seed(3)
+        model = torch.nn.Linear(4, 1)
+        opt = hvd.DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1),
+            named_parameters=model.named_parameters(),
+            backward_passes_per_step=2)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(50 + rank)
+        for step in range(2):
+            opt.zero_grad()
+            for micro in range(2):
+                loss = model(torch.randn(4, 4)).sum()
+                loss.backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        g = hvd.allgather(flat.unsqueeze(0), name="bpps")
+        assert torch.allclose(g[0], g[1], atol=1e-6)
+    """)
+
+
+def test_compression_fp16_np2():
+    run_workers(2, """
+        from horovod_amd.torch.compression import Compression
+        t = torch.randn(1000) * (rank + 1)
+        out = hvd.allreduce(t, average=False, compression=Compression.fp16,
+                            name="comp")
+        ref = hvd.allreduce(t, average=False, name="nocomp")
+        assert torch.allclose(out, ref, rtol=1e-2, atol=1e-2), \
+            (out - ref).abs().max()
+    """)
+
+
+def test_broadcast_object_np2():
+    run_workers(2, """
+        from horovod_amd.torch import broadcast_object, allgather_object
+        obj = {"a": 1, "b": [1, 2, 3]} if rank == 0 else None
+        got = broadcast_object(obj, root_rank=0)
+        assert got == {"a": 1, "b": [1, 2, 3]}, got
+        objs = allgather_object({"rank": rank})
+        assert objs == [{"rank": 0}, {"rank": 1}], objs
+    """)
+
+
+def test_broadcast_optimizer_state_np2():
+    run_workers(2, """
+        torch.manual_seed(10 + rank)  # deliberately different init
+        model = torch.nn.Linear(5, 3)
+        opt = torch.optim.Adam(model.parameters(), lr=0.01)
+        # build some state
+        loss = model(torch.randn(2, 5)).sum()
+        loss.backward()
+        opt.step()
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        from horovod_amd.torch import broadcast_optimizer_state
+        broadcast_optimizer_state(opt, root_rank=0)
+        # verify state equality via allgather of exp_avg
+        s = opt.state[list(model.parameters())[0]]["exp_avg"].flatten()
+        g = hvd.allgather(s.unsqueeze(0), name="st")
+        assert torch.allclose(g[0], g[1], atol=1e-7)
+    """)
+
+
+def test_sync_batch_norm_np2():
+    """SyncBatchNorm over 2 ranks must equal BatchNorm over the concatenated
+    batch (reference: sync_batch_norm tests)."""
+    run_workers(2, """
+        torch.manual_seed(5)
+        full = torch.randn(8, 3, 4, 4)
+        mine = full[rank * 4:(rank + 1) * 4].clone().requires_grad_(True)
+
+        sbn = hvd.SyncBatchNorm(3, momentum=0.3)
+        bn = torch.nn.BatchNorm2d(3, momentum=0.3)
+        out = sbn(mine)
+        ref_in = full.clone().requires_grad_(True)
+        ref = bn(ref_in)
+        assert torch.allclose(out, ref[rank * 4:(rank + 1) * 4], atol=1e-5), \
+            (out - ref[rank * 4:(rank + 1) * 4]).abs().max()
+        assert torch.allclose(sbn.running_mean, bn.running_mean, atol=1e-5)
+        assert torch.allclose(sbn.running_var, bn.running_var, atol=1e-4)
+        # backward equivalence
+        gout = torch.randn_like(full)
+        out.backward(gout[rank * 4:(rank + 1) * 4])
+        ref.backward(gout)
+        assert torch.allclose(mine.grad,
+                              ref_in.grad[rank * 4:(rank + 1) * 4],
+                              atol=1e-4), \
+            (mine.grad - ref_in.grad[rank*4:(rank+1)*4]).abs().max()
+    """)
+
+
+def _adasum_golden(tensors):
+    """NumPy golden model of the VHDD combine tree (matches core.cc and
+    adasum_kernels.hip; reference numerics: adasum.h:396-412)."""
+    def combine(a, b):
+        dot = float(np.dot(a, b))
+        na = float(np.dot(a, a))
+        nb = float(np.dot(b, b))
+        ac = 1.0 - dot / (2 * na) if na > 0 else 1.0
+        bc = 1.0 - dot / (2 * nb) if nb > 0 else 1.0
+        return ac * a + bc * b
+
+    work = [t.astype(np.float64) for t in tensors]
+    p = 1
+    while p * 2 <= len(work):
+        p *= 2
+    for i in range(p, len(work)):
+        work[i - p] = combine(work[i - p], work[i])
+    work = work[:p]
+    stride = 1
+    while stride < p:
+        i = 0
+        while i + stride < p:
+            work[i] = combine(work[i], work[i + stride])
+            i += 2 * stride
+        stride *= 2
+    return work[0]
+
+
+def test_adasum_golden_np2():
+    rng = np.random.RandomState(42)
+    a = rng.randn(64).astype(np.float32)
+    b = rng.randn(64).astype(np.float32)
+    expected = _adasum_golden([a, b])
+    run_workers(2, f"""
+        import numpy as np
+        vecs = [np.array({a.tolist()!r}, dtype=np.float32),
+                np.array({b.tolist()!r}, dtype=np.float32)]
+        t = torch.from_numpy(vecs[rank]).clone()
+        out = hvd.allreduce(t, op=hvd.Adasum, name="adasum")
+        expected = np.array({expected.tolist()!r})
+        assert np.allclose(out.numpy(), expected, atol=1e-5), \\
+            np.abs(out.numpy() - expected).max()
+    """)
+
+
+def test_adasum_golden_np3():
+    rng = np.random.RandomState(7)
+    vecs = [rng.randn(33).astype(np.float32) for _ in range(3)]
+    expected = _adasum_golden(vecs)
+    run_workers(3, f"""
+        import numpy as np
+        allv = {[v.tolist() for v in vecs]!r}
+        t = torch.tensor(allv[rank], dtype=torch.float32)
+        out = hvd.allreduce(t, op=hvd.Adasum, name="adasum3")
+        expected = np.array({expected.tolist()!r})
+        assert np.allclose(out.numpy(), expected, atol=1e-5), \\
+            np.abs(out.numpy() - expected).max()
+    """)
+
+
+def test_autograd_allgather_np2():
+    run_workers(2, """
+        x = (torch.arange(4).float() + rank).requires_grad_(True)
+        y = hvd.allgather(x, name="ag_grad")
+        loss = (y * torch.arange(8).float()).sum()
+        loss.backward()
+        expected = torch.arange(rank * 4, rank * 4 + 4).float() * 2
+        assert torch.allclose(x.grad, expected), (x.grad, expected)
+    """)
+
+
+def test_autograd_broadcast_np2():
+    run_workers(2, """
+        x = torch.ones(3, requires_grad=True)
+        y = hvd.broadcast(x, root_rank=0, name="bc_grad")
+        y.sum().backward()
+        if rank == 0:
+            assert torch.allclose(x.grad, torch.full((3,), 2.0)), x.grad
+        else:
+            assert torch.allclose(x.grad, torch.zeros(3)), x.grad
+    """)
+
+
+def test_sparse_allreduce_np2():
+    run_workers(2, """
+        i = torch.tensor([[0, 2]]) if rank == 0 else torch.tensor([[1, 2]])
+        v = torch.tensor([1.0, 2.0]) if rank == 0 else torch.tensor([3.0, 4.0])
+        sp = torch.sparse_coo_tensor(i, v, (4,))
+        closure = hvd.sparse_allreduce_async(sp, name="sp", op=hvd.Sum)
+        out = closure().to_dense()
+        assert torch.allclose(out, torch.tensor([1.0, 3.0, 6.0, 0.0])), out
+    """)
