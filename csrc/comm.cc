@@ -5,6 +5,7 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
+#include <poll.h>
 #include <unistd.h>
 
 #include <chrono>
@@ -54,6 +55,16 @@ void StarComm::Init(int rank, int size, const std::string& addr, int port,
     if (listen(listen_fd_, size_) != 0) comm_error("listen");
     int connected = 0;
     while (connected < size_ - 1) {
+      // bounded accept: poll with the remaining init deadline so a missing
+      // worker turns into an error instead of an eternal hang
+      auto remain = std::chrono::duration<double>(
+                        deadline - std::chrono::steady_clock::now())
+                        .count();
+      if (remain <= 0) comm_error("accept timeout (worker never connected)");
+      struct pollfd pfd{listen_fd_, POLLIN, 0};
+      int pr = poll(&pfd, 1, (int)(remain * 1000));
+      if (pr == 0) comm_error("accept timeout (worker never connected)");
+      if (pr < 0) comm_error("poll");
       int fd = accept(listen_fd_, nullptr, nullptr);
       if (fd < 0) comm_error("accept");
       set_nodelay(fd);

@@ -571,8 +571,23 @@ void Abort(GlobalState& st, const std::string& why) {
 }
 
 void BackgroundLoop(GlobalState& st) {
+  // Shutdown protocol: a graceful exit needs every rank's agreement (AND bit
+  // in the cache-sync round) so no peer is left mid-collective; but a rank
+  // must never hang forever on a dead peer — after the grace window it
+  // leaves unilaterally and survivors see a socket error (-> their pending
+  // ops fail with HorovodInternalError, the elastic recovery signal).
+  const char* grace_env = std::getenv("HOROVOD_SHUTDOWN_GRACE_SECONDS");
+  double grace = grace_env ? atof(grace_env) : 5.0;
+  std::chrono::steady_clock::time_point shutdown_since{};
   while (true) {
     auto cycle_start = std::chrono::steady_clock::now();
+    if (st.shutdown_requested) {
+      if (shutdown_since.time_since_epoch().count() == 0)
+        shutdown_since = cycle_start;
+      else if (std::chrono::duration<double>(cycle_start - shutdown_since)
+                   .count() > grace)
+        break;
+    }
     auto reqs = st.queue.PopMessages();
     ResponseList rl;
     try {
