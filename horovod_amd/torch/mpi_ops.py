@@ -18,7 +18,20 @@ from horovod_amd.common.process_sets import (ProcessSet, global_process_set,
                                              add_process_set,  # noqa: F401
                                              remove_process_set)  # noqa: F401
 
+from horovod_amd.common.exceptions import HorovodInternalError
+
 _basics = HorovodBasics()
+
+
+def _translate_error(fn, *args, **kwargs):
+    """Native-core errors tagged HorovodInternalError become the typed
+    exception the elastic retry loop catches."""
+    try:
+        return fn(*args, **kwargs)
+    except RuntimeError as e:
+        if "HorovodInternalError" in str(e):
+            raise HorovodInternalError(str(e)) from None
+        raise
 
 init = _basics.init
 shutdown = _basics.shutdown
@@ -134,7 +147,7 @@ def poll(handle):
 def synchronize(handle):
     """Wait for the async op and return its output tensor(s)."""
     info = _handles.pop(handle, None)
-    outs, extra, result_int = _core.wait(handle)
+    outs, extra, result_int = _translate_error(_core.wait, handle)
     if info is not None and info.kind == "join":
         return result_int
     if info is not None and info.kind == "alltoall_splits":

@@ -214,6 +214,16 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._synchronized = False
         return super(self.__class__, self).step(closure)
 
+    def reset_distributed_state(self):
+        """Drop in-flight allreduce bookkeeping after an elastic reset (the
+        native handles died with the old controller)."""
+        self._handles.clear()
+        for p in self._allreduce_delay:
+            self._allreduce_delay[p] = self.backward_passes_per_step
+        for gi in self._group_counts:
+            self._group_counts[gi] = 0
+        self._synchronized = False
+
     def zero_grad(self, *args, **kwargs):
         if self._handles:
             raise AssertionError(

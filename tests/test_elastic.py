@@ -1,0 +1,164 @@
+"""Elastic tests (reference model: test/single/test_elastic_driver.py +
+test/integration/test_elastic_torch.py — driver logic with controlled
+discovery, and real worker processes driven through scale-up/down and
+failure)."""
+import os
+import subprocess
+import sys
+import textwrap
+import time
+
+import pytest
+import torch
+
+from tests.parallel_util import REPO
+
+
+WORKER = """
+import os, sys, time
+import torch
+import horovod_amd.torch as hvd
+import horovod_amd.torch.elastic as elastic
+
+hvd.init()
+torch.manual_seed(0)
+model = torch.nn.Linear(4, 1)
+opt = torch.optim.SGD(model.parameters(), lr=0.01)
+opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters())
+state = elastic.TorchState(model, opt, batch=0)
+
+TARGET = int(os.environ.get("TEST_TARGET_BATCHES", "30"))
+CRASH_AT = int(os.environ.get("TEST_CRASH_AT", "-1"))
+CRASH_WID = os.environ.get("TEST_CRASH_WID", "")
+MARKER = os.environ.get("TEST_MARKER_FILE", "")
+
+@elastic.run
+def train(state):
+    while state.batch < TARGET:
+        if (state.batch == CRASH_AT and
+                os.environ.get("HOROVOD_WORKER_ID") == CRASH_WID and
+                not os.path.exists(MARKER + ".crashed")):
+            open(MARKER + ".crashed", "w").write("x")
+            os._exit(17)
+        time.sleep(0.03)
+        opt.zero_grad()
+        loss = model(torch.ones(2, 4)).sum()
+        loss.backward()
+        opt.step()
+        state.batch += 1
+        state.commit()
+
+train(state)
+with open(MARKER, "a") as f:
+    f.write(f"done rank={hvd.rank()} size={hvd.size()} batch={state.batch}\\n")
+"""
+
+
+def _driver(tmp_path, hosts, extra_env=None, **kw):
+    from horovod_amd.runner.elastic_driver import ElasticDriver, FixedHosts
+    marker = str(tmp_path / "marker.txt")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["TEST_MARKER_FILE"] = marker
+    env["HOROVOD_SHUTDOWN_GRACE_SECONDS"] = "2"
+    env.update(extra_env or {})
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER)
+    discovery = FixedHosts(hosts)
+    driver = ElasticDriver(discovery, [sys.executable, str(script)], env=env,
+                           cooldown=kw.pop("cooldown", 0.0), **kw)
+    return driver, discovery, marker
+
+
+def test_elastic_completes_static(tmp_path):
+    driver, disc, marker = _driver(tmp_path, {"127.0.0.1": 2})
+    driver.start()
+    err = driver.wait_for_result(timeout=120)
+    driver.stop()
+    assert err is None, err
+    text = open(marker).read()
+    assert "size=2" in text and "batch=30" in text, text
+
+
+def test_elastic_worker_failure_recovery(tmp_path):
+    driver, disc, marker = _driver(
+        tmp_path, {"127.0.0.1": 2},
+        extra_env={"TEST_CRASH_AT": "5", "TEST_CRASH_WID": "127.0.0.1:1"})
+    driver.start()
+    err = driver.wait_for_result(timeout=180)
+    driver.stop()
+    assert err is None, err
+    text = open(marker).read()
+    # survivor restored from commit and finished (a respawned peer may also
+    # finish; at least one rank reaches the target)
+    assert "batch=30" in text, text
+
+
+def test_elastic_scale_up(tmp_path):
+    driver, disc, marker = _driver(
+        tmp_path, {"127.0.0.1": 1},
+        extra_env={"TEST_TARGET_BATCHES": "60"})
+    driver.start()
+    time.sleep(2.0)
+    disc.set({"127.0.0.1": 2})
+    err = driver.wait_for_result(timeout=180)
+    driver.stop()
+    assert err is None, err
+    text = open(marker).read()
+    assert "size=2" in text, text
+
+
+def test_elastic_min_np_violation(tmp_path):
+    driver, disc, marker = _driver(tmp_path, {"127.0.0.1": 2}, min_np=2,
+                                   extra_env={"TEST_TARGET_BATCHES": "200"})
+    driver.start()
+    time.sleep(1.5)
+    disc.set({"127.0.0.1": 1})
+    err = driver.wait_for_result(timeout=120)
+    driver.stop()
+    assert err is not None
+
+
+# ---------------------------------------------------------------------------
+# unit tests (no cluster)
+# ---------------------------------------------------------------------------
+
+def test_elastic_sampler_unit():
+    from horovod_amd.torch.elastic.sampler import ElasticSampler
+
+    class DS:
+        def __len__(self):
+            return 10
+
+    s = ElasticSampler(DS(), shuffle=False)
+    assert len(list(iter(s))) == 10
+    # record first 4 as processed, reset -> 6 remain
+    s.processed_indices.update(list(iter(s))[:4])
+    s.reset()
+    assert len(list(iter(s))) == 6
+    state = s.state_dict()
+    s2 = ElasticSampler(DS(), shuffle=False)
+    s2.load_state_dict(state)
+    s2.reset()
+    assert len(list(iter(s2))) == 6
+    s.set_epoch(1)
+    assert len(list(iter(s))) == 10
+
+
+def test_torch_state_save_restore_unit():
+    import horovod_amd.torch as hvd
+    hvd.init()  # single-process
+    from horovod_amd.torch.elastic.state import TorchState
+    model = torch.nn.Linear(3, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    state = TorchState(model, opt, epoch=5, batch=7)
+    state.save()
+    before = [p.detach().clone() for p in model.parameters()]
+    with torch.no_grad():
+        for p in model.parameters():
+            p.add_(1.0)
+    state.epoch = 9
+    state.restore()
+    for p, b in zip(model.parameters(), before):
+        assert torch.allclose(p, b)
+    assert state.epoch == 5
