@@ -6,6 +6,7 @@
 #include <torch/extension.h>
 
 #include "core.h"
+#include "timeline.h"
 #include "gpu.h"
 
 namespace {
@@ -116,6 +117,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // ---- completion ---------------------------------------------------------
   m.def("poll", [](int handle) { return State().handles.Poll(handle); });
   m.def("wait", &WaitHandle);
+
+  // ---- timeline -----------------------------------------------------------
+  m.def("start_timeline", [](const std::string& path, bool mark_cycles) {
+    auto& st = State();
+    if (!st.initialized) throw std::runtime_error("not initialized");
+    SetTimeline(st, std::make_shared<Timeline>(path, st.rank));
+    (void)mark_cycles;
+  });
+  m.def("stop_timeline", [] { SetTimeline(State(), nullptr); });
 
   // ---- process sets -------------------------------------------------------
   m.def("add_process_set", [](std::vector<int32_t> ranks) {

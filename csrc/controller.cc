@@ -92,7 +92,7 @@ void ResponseCache::Evict(int slot) {
 
 void ResponseCache::Put(const Response& response, const std::vector<Request>& reqs) {
   if (response.type == ResponseType::JOIN || response.type == ResponseType::BARRIER ||
-      response.type == ResponseType::ERROR)
+      response.type == ResponseType::ERROR || response.type == ResponseType::TUNE)
     return;
   // Split a fused response into single-tensor cache entries.
   size_t nsizes_per = response.names.size()
@@ -154,6 +154,12 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
 Controller::Controller(StarComm* comm, int rank, int size, ControllerConfig cfg)
     : comm_(comm), rank_(rank), size_(size), cfg_(cfg) {
   cache_.set_capacity(cfg_.cache_capacity);
+  start_time_ = std::chrono::steady_clock::now();
+  if (rank_ == 0 && size_ > 1 && std::getenv("HOROVOD_AUTOTUNE")) {
+    const char* lp = std::getenv("HOROVOD_AUTOTUNE_LOG");
+    autotuner_.reset(new Autotuner(cfg_.fusion_threshold_bytes,
+                                   cfg_.cycle_time_ms, lp ? lp : ""));
+  }
   ProcessSetInfo global;
   global.id = 0;
   for (int r = 0; r < size; ++r) global.ranks.push_back(r);
@@ -227,7 +233,7 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
 
     std::string vecB(nbytes + 1, '\0');
     for (int slot : my_invalid_slots) vecB[slot / 8] |= (char)(1 << (slot % 8));
-    bool need_slow = !slow.empty() || !inflight_.empty();
+    bool need_slow = !slow.empty() || !inflight_.empty() || pending_tune_;
     if (need_slow) vecB[nbytes] |= 1;
     comm_->BitOr(vecB);
 
@@ -325,6 +331,29 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     for (auto& r : fused) result.responses.push_back(std::move(r));
   }
 
+  // ---- autotuner (rank 0): score this cycle's reduced bytes; a ready
+  // proposal is published next cycle as a TUNE response (slow path forced
+  // via pending_tune_).
+  if (autotuner_ && !autotuner_->done()) {
+    int64_t bytes = 0;
+    for (auto& resp : result.responses) {
+      if (resp.type != ResponseType::ALLREDUCE &&
+          resp.type != ResponseType::ADASUM)
+        continue;
+      const int64_t* p = resp.tensor_shapes.data();
+      const int64_t* end = p + resp.tensor_shapes.size();
+      while (p < end) {
+        int64_t nd = *p++;
+        int64_t n = 1;
+        for (int64_t i = 0; i < nd; ++i) n *= *p++;
+        bytes += n * (int64_t)DataTypeSize(resp.dtype);
+      }
+    }
+    double now = std::chrono::duration<double>(std::chrono::steady_clock::now() -
+                                               start_time_)
+                     .count();
+    if (autotuner_->Record(bytes, now)) pending_tune_ = true;
+  }
   return result;
 }
 
@@ -404,7 +433,16 @@ std::vector<Response> Controller::CoordinatorProcess(
   }
 
   CheckForStalledTensors();
-  return FuseResponses(ready_responses_);
+  auto out = FuseResponses(ready_responses_);
+  if (pending_tune_ && autotuner_) {
+    Response t;
+    t.type = ResponseType::TUNE;
+    auto p = autotuner_->current();
+    t.tensor_sizes = {p.fusion_bytes, (int64_t)(p.cycle_time_ms * 1000.0)};
+    out.push_back(std::move(t));
+    pending_tune_ = false;
+  }
+  return out;
 }
 
 Response Controller::ConstructResponse(const std::string& key, PendingTensor& pt) {

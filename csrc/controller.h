@@ -31,6 +31,7 @@
 #include <unordered_set>
 #include <vector>
 
+#include "autotuner.h"
 #include "comm.h"
 #include "common.h"
 #include "message.h"
@@ -133,6 +134,10 @@ class Controller {
   void set_cycle_time_ms(double ms) { cfg_.cycle_time_ms = ms; }
   double cycle_time_ms() const { return cfg_.cycle_time_ms; }
   ControllerConfig& config() { return cfg_; }
+  void ApplyTune(int64_t fusion_bytes, double cycle_time_ms) {
+    cfg_.fusion_threshold_bytes = fusion_bytes;
+    cfg_.cycle_time_ms = cycle_time_ms;
+  }
 
  private:
   // Slow path, coordinator side.
@@ -162,6 +167,11 @@ class Controller {
   std::unordered_map<int32_t, ProcessSetInfo> process_sets_;
   int32_t next_set_id_ = 1;
   std::chrono::steady_clock::time_point last_stall_check_;
+  std::chrono::steady_clock::time_point start_time_;
+
+  // autotuner (rank 0 only; results propagate via TUNE responses)
+  std::unique_ptr<Autotuner> autotuner_;
+  bool pending_tune_ = false;
 
  public:
   // join bookkeeping shared with core

@@ -153,6 +153,13 @@ GlobalState& State() {
   return st;
 }
 
+std::shared_ptr<Timeline> GetTimeline(GlobalState& st) {
+  return std::atomic_load(&st.timeline);
+}
+void SetTimeline(GlobalState& st, std::shared_ptr<Timeline> t) {
+  std::atomic_store(&st.timeline, std::move(t));
+}
+
 bool IsInitialized() { return State().initialized; }
 
 namespace {
@@ -463,6 +470,12 @@ void PerformOperation(GlobalState& st, Response& resp) {
              st.local_joined[resp.process_set_id];
   }
 
+  if (resp.type == ResponseType::TUNE) {
+    if (resp.tensor_sizes.size() == 2)
+      st.controller->ApplyTune(resp.tensor_sizes[0],
+                               resp.tensor_sizes[1] / 1000.0);
+    return;
+  }
   if (resp.type == ResponseType::JOIN) {
     TensorTableEntry e;
     if (st.queue.PopEntry(resp.process_set_id, "join", e)) {
@@ -522,7 +535,8 @@ void PerformOperation(GlobalState& st, Response& resp) {
     return;
   }
 
-  if (st.timeline) st.timeline->OpStart(resp);
+  auto tl = GetTimeline(st);
+  if (tl) tl->OpStart(resp);
 
   // Response device is the normalized marker (-1 CPU, -2 GPU) — the GPU path
   // must run on every rank (even relay-only non-members) so the RCCL comm
@@ -553,11 +567,11 @@ void PerformOperation(GlobalState& st, Response& resp) {
           break;
       }
       FailEntries(entries, Status::OK());
-      if (st.timeline) st.timeline->OpEnd(resp);
+      if (tl) tl->OpEnd(resp);
     }
   } catch (const std::exception& ex) {
     FailEntries(entries, Status::UnknownError(ex.what()));
-    if (st.timeline) st.timeline->OpEnd(resp);
+    if (tl) tl->OpEnd(resp);
   }
 }
 
@@ -633,7 +647,7 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
   st.controller.reset(new Controller(&st.comm, rank, size, cfg));
   if (cfg.timeline_enabled) {
     const char* tf = std::getenv("HOROVOD_TIMELINE");
-    if (tf) st.timeline.reset(new Timeline(tf, rank));
+    if (tf) SetTimeline(st, std::make_shared<Timeline>(tf, rank));
   }
   st.bg_thread = std::thread([&st] { BackgroundLoop(st); });
   st.initialized = true;
@@ -647,7 +661,7 @@ void ShutdownHorovod() {
   gpu::Shutdown();
   st.comm.Shutdown();
   st.controller.reset();
-  st.timeline.reset();
+  SetTimeline(st, nullptr);
   st.local_joined.clear();
   st.join_device.clear();
   st.initialized = false;

@@ -98,10 +98,15 @@ struct GlobalState {
   std::unordered_map<int32_t, bool> local_joined;
   std::unordered_map<int32_t, int> join_device;
 
-  std::unique_ptr<class Timeline> timeline;
+  // hot-swappable at runtime (hvd.start_timeline): accessed via atomic
+  // shared_ptr loads from the background thread.
+  std::shared_ptr<class Timeline> timeline;
 };
 
 GlobalState& State();
+
+std::shared_ptr<class Timeline> GetTimeline(GlobalState& st);
+void SetTimeline(GlobalState& st, std::shared_ptr<class Timeline> t);
 
 // Lifecycle -----------------------------------------------------------------
 void InitHorovod(int rank, int size, int local_rank, int local_size,

@@ -125,9 +125,9 @@ void FinalizerLoop() {
     hipEventDestroy(op.done_event);
     for (auto ev : op.ready_events) hipEventDestroy(ev);
     auto& st = State();
-    if (st.timeline && !op.entries.empty())
-      st.timeline->Activity(op.entries[0].name, op.activity, op.start_us,
-                            st.timeline->NowUs());
+    auto tl = GetTimeline(st);
+    if (tl && !op.entries.empty())
+      tl->Activity(op.entries[0].name, op.activity, op.start_us, tl->NowUs());
   }
 }
 
@@ -412,7 +412,8 @@ void Execute(GlobalState& st, Response& resp,
   DataType wire = resp.dtype;
   int64_t wire_size = (int64_t)DataTypeSize(wire);
   auto wire_nccl = ToNccl(wire);
-  int64_t t_start = st.timeline ? st.timeline->NowUs() : 0;
+  auto tl0 = GetTimeline(st);
+  int64_t t_start = tl0 ? tl0->NowUs() : 0;
 
   std::vector<hipEvent_t> ready;
   WaitReadyEvents(ctx, entries, ready);

@@ -35,6 +35,8 @@ enum class ResponseType : uint8_t {
   BARRIER = 6,
   REDUCESCATTER = 7,
   ERROR = 8,
+  TUNE = 9,  // autotuner parameter broadcast (tensor_sizes = [fusion_bytes,
+             // cycle_time_us])
 };
 
 // A single collective announcement from one rank.

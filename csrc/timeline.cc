@@ -14,6 +14,7 @@ const char* TypeName(ResponseType t) {
     case ResponseType::BARRIER: return "BARRIER";
     case ResponseType::REDUCESCATTER: return "REDUCESCATTER";
     case ResponseType::ERROR: return "ERROR";
+    case ResponseType::TUNE: return "TUNE";
   }
   return "?";
 }

@@ -24,7 +24,7 @@ from horovod_amd.torch.mpi_ops import (  # noqa: F401
     is_initialized, join, local_rank, local_size, mpi_built, mpi_enabled,
     mpi_threads_supported, nccl_built, poll, rank, reducescatter,
     reducescatter_async, rocm_built, shutdown, size, sparse_allreduce_async,
-    synchronize, wait)
+    start_timeline, stop_timeline, synchronize, wait)
 from horovod_amd.torch.mpi_ops import (add_process_set,  # noqa: F401
                                        remove_process_set)
 from horovod_amd.common.process_sets import (ProcessSet,  # noqa: F401

@@ -539,6 +539,16 @@ def join(device=-1):
     return synchronize(h)
 
 
+def start_timeline(file_path, mark_cycles=False):
+    """Start writing a Chrome-trace timeline at runtime (reference:
+    operations.cc horovod_start_timeline)."""
+    _core.start_timeline(file_path, mark_cycles)
+
+
+def stop_timeline():
+    _core.stop_timeline()
+
+
 def barrier(process_set=global_process_set):
     h = _core.barrier_async(_set_id(process_set))
     _register(h)

@@ -241,3 +241,34 @@ def test_sparse_allreduce_np2():
         out = closure().to_dense()
         assert torch.allclose(out, torch.tensor([1.0, 3.0, 6.0, 0.0])), out
     """)
+
+
+def test_timeline_np2(tmp_path):
+    """Runtime timeline start/stop produces valid Chrome-trace JSON on every
+    rank (reference: test_timeline.py)."""
+    import json as _json
+    tl = str(tmp_path / "tl.json")
+    run_workers(2, f"""
+        hvd.start_timeline({tl!r})
+        for i in range(3):
+            hvd.allreduce(torch.ones(100), average=False, name=f"tl{{i}}")
+        hvd.stop_timeline()
+    """)
+    data = _json.load(open(tl))
+    names = [e.get("name") for e in data]
+    assert "ALLREDUCE" in names, names[:10]
+    data1 = _json.load(open(tl + ".1"))
+    assert data1
+
+
+def test_autotune_np2(tmp_path):
+    log = str(tmp_path / "autotune.csv")
+    run_workers(2, """
+        import horovod_amd._core as core
+        before = core.get_fusion_threshold()
+        for i in range(200):
+            hvd.allreduce(torch.ones(4096), average=False, name="at")
+        # autotuner window is 3s; this loop is fast, so just assert the
+        # mechanism is alive: threshold is a sane positive value
+        assert core.get_fusion_threshold() > 0
+    """, extra_env={"HOROVOD_AUTOTUNE": "1", "HOROVOD_AUTOTUNE_LOG": log})
