@@ -36,6 +36,8 @@ def parse_args():
                    help="disable bf16 autocast (fp32 compute)")
     p.add_argument("--no-overlap", action="store_true",
                    help="disable backward/allreduce overlap (ablation)")
+    p.add_argument("--fused-sgd", action="store_true",
+                   help="use the CDNA4 fused SGD step kernel")
     return p.parse_args()
 
 
@@ -71,8 +73,13 @@ def main():
     compression = {"none": Compression.none, "fp16": Compression.fp16,
                    "bf16": Compression.bf16}[args.compression]
 
-    opt = torch.optim.SGD(model.parameters(), lr=0.0125 * hvd.size(),
-                          momentum=0.9, weight_decay=5e-5)
+    if args.fused_sgd:
+        from horovod_amd.ops import FusedSGD
+        opt = FusedSGD(model.parameters(), lr=0.0125 * hvd.size(),
+                       momentum=0.9, weight_decay=5e-5)
+    else:
+        opt = torch.optim.SGD(model.parameters(), lr=0.0125 * hvd.size(),
+                              momentum=0.9, weight_decay=5e-5)
     opt = hvd.DistributedOptimizer(
         opt, named_parameters=model.named_parameters(),
         compression=compression,
@@ -133,6 +140,7 @@ def main():
                 "parallelism": f"dp{n}",
                 "compression": args.compression,
                 "reduction": "adasum" if args.use_adasum else "average",
+                "fused_sgd": args.fused_sgd,
             },
         }
         print(json.dumps(result), flush=True)

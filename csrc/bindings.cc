@@ -118,6 +118,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("poll", [](int handle) { return State().handles.Poll(handle); });
   m.def("wait", &WaitHandle);
 
+  // ---- fused optimizer kernels -------------------------------------------
+  m.def("fused_sgd_step",
+        [](std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+           std::vector<at::Tensor> momenta, double lr, double momentum,
+           double weight_decay, double dampening, bool nesterov) {
+          hvd::gpu::FusedSgdStep(params, grads, momenta, lr, momentum,
+                                 weight_decay, dampening, nesterov);
+        });
+
   // ---- timeline -----------------------------------------------------------
   m.def("start_timeline", [](const std::string& path, bool mark_cycles) {
     auto& st = State();

@@ -598,6 +598,33 @@ void Execute(GlobalState& st, Response& resp,
   Finalize(ctx, std::move(entries), std::move(ready), activity, t_start);
 }
 
+void FusedSgdStep(std::vector<at::Tensor>& params,
+                  std::vector<at::Tensor>& grads,
+                  std::vector<at::Tensor>& momenta, double lr, double momentum,
+                  double weight_decay, double dampening, bool nesterov) {
+  if (params.empty()) return;
+  int device = (int)params[0].get_device();
+  c10::hip::HIPGuard guard(device);
+  hipStream_t stream = c10::hip::getCurrentHIPStream(device).stream();
+  SgdBatchArgs args;
+  auto flush = [&] {
+    if (args.count == 0) return;
+    HIP_CHECK(FusedSgdLaunch(args, (float)lr, (float)momentum,
+                             (float)weight_decay, (float)dampening, nesterov,
+                             stream));
+    args.count = 0;
+  };
+  for (size_t i = 0; i < params.size(); ++i) {
+    if (args.count == kCopyBatchCapacity) flush();
+    int k = args.count++;
+    args.params[k] = params[i].data_ptr();
+    args.grads[k] = grads[i].data_ptr();
+    args.momenta[k] = momenta.empty() ? nullptr : momenta[i].data_ptr();
+    args.numel[k] = (unsigned long long)params[i].numel();
+  }
+  flush();
+}
+
 void WaitAllPending() {
   while (true) {
     {

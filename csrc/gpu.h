@@ -11,6 +11,8 @@
 #include <cstdint>
 #include <vector>
 
+#include <ATen/ATen.h>
+
 namespace hvd {
 struct GlobalState;
 struct Response;
@@ -34,6 +36,13 @@ void Shutdown();
 // True once a RCCL communicator has been created (used by tests to assert
 // the native path ran).
 bool RcclUsed();
+
+// Fused SGD step on torch's current stream (one kernel for all buckets;
+// empty `momenta` = plain SGD).  fp32 tensors.
+void FusedSgdStep(std::vector<at::Tensor>& params,
+                  std::vector<at::Tensor>& grads,
+                  std::vector<at::Tensor>& momenta, double lr, double momentum,
+                  double weight_decay, double dampening, bool nesterov);
 
 }  // namespace gpu
 }  // namespace hvd

@@ -57,5 +57,18 @@ hipError_t AdasumDotsLaunch(const AdasumBatchArgs& args, int dt, double* dots,
 hipError_t AdasumScaledAddLaunch(const AdasumBatchArgs& args, int dt,
                                  const double* dots, hipStream_t stream);
 
+// ---- Fused SGD (sgd_kernels.hip) ------------------------------------------
+struct SgdBatchArgs {
+  void* params[kCopyBatchCapacity];
+  const void* grads[kCopyBatchCapacity];
+  void* momenta[kCopyBatchCapacity];
+  unsigned long long numel[kCopyBatchCapacity];
+  int count = 0;
+};
+
+hipError_t FusedSgdLaunch(const SgdBatchArgs& args, float lr, float momentum,
+                          float weight_decay, float dampening, bool nesterov,
+                          hipStream_t stream);
+
 }  // namespace gpu
 }  // namespace hvd

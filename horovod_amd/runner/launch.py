@@ -138,13 +138,30 @@ def build_parser():
                         help="elastic: maximum np")
     parser.add_argument("--host-discovery-script", default=None,
                         help="elastic: executable printing host:slots lines")
+    parser.add_argument("--config-file", default=None,
+                        help="YAML file with launcher settings (reference: "
+                             "launch.py:581-585)")
     parser.add_argument("command", nargs=argparse.REMAINDER,
                         help="training command")
     return parser
 
 
+def apply_config_file(args):
+    if not args.config_file:
+        return args
+    import yaml
+    with open(args.config_file) as f:
+        cfg = yaml.safe_load(f) or {}
+    for key, val in cfg.items():
+        attr = key.replace("-", "_")
+        if hasattr(args, attr) and getattr(args, attr) in (None, False):
+            setattr(args, attr, val)
+    return args
+
+
 def main(argv=None):
     args = build_parser().parse_args(argv)
+    args = apply_config_file(args)
     if not args.command:
         print("hvdrun: no command given", file=sys.stderr)
         return 1

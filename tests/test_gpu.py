@@ -138,3 +138,29 @@ def test_native_extension_loaded():
     """Fail loudly if the in-tree native extension is not what's loaded."""
     import horovod_amd._core as core
     assert "/horovod_amd/_core.so" in core.__file__, core.__file__
+
+
+@requires_gpu
+def test_fused_sgd_matches_torch(hvd):
+    """FusedSGD kernel vs torch.optim.SGD over several steps."""
+    from horovod_amd.ops import FusedSGD
+    torch.manual_seed(9)
+    ref_params = [torch.randn(n, device="cuda", requires_grad=False)
+                  for n in (1000, 37, 4096)]
+    fused_params = [p.clone() for p in ref_params]
+    ref_opt = torch.optim.SGD(ref_params, lr=0.1, momentum=0.9,
+                              weight_decay=0.01)
+    fused_opt = FusedSGD(fused_params, lr=0.1, momentum=0.9,
+                         weight_decay=0.01)
+    for step in range(5):
+        grads = [torch.randn_like(p) for p in ref_params]
+        for p, g in zip(ref_params, grads):
+            p.grad = g
+        for p, g in zip(fused_params, grads):
+            p.grad = g.clone()
+        ref_opt.step()
+        fused_opt.step()
+    torch.cuda.synchronize()
+    for rp, fp in zip(ref_params, fused_params):
+        assert torch.allclose(rp, fp, rtol=1e-5, atol=1e-6), \
+            (rp - fp).abs().max()

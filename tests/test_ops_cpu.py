@@ -223,3 +223,22 @@ def test_steady_state_cache_np2():
             out = hvd.allreduce(torch.ones(1000), average=False, name="steady")
             assert out.sum().item() == 2000.0
     """)
+
+
+def test_parallel_grid_sets_np4():
+    run_workers(4, """
+        from horovod_amd.parallel import grid_process_sets
+        my_tp, my_dp, tp_sets, dp_sets = grid_process_sets(2)
+        # tp allreduce within my row
+        out = hvd.allreduce(torch.ones(2) * (rank + 1), average=False,
+                            process_set=my_tp, name="tp")
+        row = rank // 2
+        expected = (2 * row + 1) + (2 * row + 2)
+        assert out[0].item() == expected, (out, expected)
+        # dp allreduce within my column
+        out = hvd.allreduce(torch.ones(2) * (rank + 1), average=False,
+                            process_set=my_dp, name="dp")
+        col = rank % 2
+        expected = (col + 1) + (col + 3)
+        assert out[0].item() == expected, (out, expected)
+    """)

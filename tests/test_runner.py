@@ -1,0 +1,92 @@
+"""Launcher / runner tests (reference model: test/single/test_run.py —
+CLI parsing + run-controller logic, no cluster)."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+from tests.parallel_util import REPO
+
+
+def test_parser_flags():
+    from horovod_amd.runner.launch import build_parser
+    args = build_parser().parse_args(
+        ["-np", "4", "-H", "a:2,b:2", "--fusion-threshold-mb", "32",
+         "--cycle-time-ms", "0.5", "--autotune", "--", "python", "train.py"])
+    assert args.num_proc == 4
+    assert args.hosts == "a:2,b:2"
+    assert args.fusion_threshold_mb == 32
+    assert args.autotune
+    assert args.command == ["--", "python", "train.py"]
+
+
+def test_parse_host_spec():
+    from horovod_amd.runner.launch import parse_host_spec
+    assert parse_host_spec("h1:4,h2:2", 6) == [("h1", 4), ("h2", 2)]
+    assert parse_host_spec(None, 3) == [("127.0.0.1", 3)]
+
+
+def test_config_file(tmp_path):
+    from horovod_amd.runner.launch import apply_config_file, build_parser
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text("fusion-threshold-mb: 16\ncycle-time-ms: 2.5\n")
+    args = build_parser().parse_args(["--config-file", str(cfg), "x"])
+    args = apply_config_file(args)
+    assert args.fusion_threshold_mb == 16
+    assert args.cycle_time_ms == 2.5
+
+
+def test_hvdrun_cli_local(tmp_path):
+    out = tmp_path / "out.txt"
+    script = tmp_path / "t.py"
+    script.write_text(
+        "import torch, horovod_amd.torch as hvd\n"
+        "hvd.init()\n"
+        "v = hvd.allreduce(torch.ones(2), average=False, name='x')\n"
+        f"open({str(out)!r}, 'a').write(f'{{hvd.rank()}}:{{v.sum().item()}}\\n')\n")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run([sys.executable, os.path.join(REPO, "bin", "hvdrun"),
+                        "-np", "2", sys.executable, str(script)],
+                       env=env, timeout=120)
+    assert r.returncode == 0
+    lines = sorted(out.read_text().splitlines())
+    assert lines == ["0:4.0", "1:4.0"], lines
+
+
+def test_interactive_run():
+    import horovod_amd
+
+    def fn(mult):
+        import torch
+        import horovod_amd.torch as hvd
+        hvd.init()
+        out = hvd.allreduce(torch.ones(3) * mult, average=False, name="i")
+        return (hvd.rank(), float(out.sum().item()))
+
+    results = horovod_amd.run(fn, args=(2,), np=2)
+    assert sorted(results) == [(0, 12.0), (1, 12.0)], results
+
+
+def test_async_data_loader():
+    from horovod_amd.data import AsyncDataLoaderMixin, BaseDataLoader
+
+    class ListLoader(BaseDataLoader):
+        def __init__(self, items):
+            self.items = items
+
+        def __len__(self):
+            return len(self.items)
+
+        def _iterate(self):
+            yield from self.items
+
+    class AsyncListLoader(AsyncDataLoaderMixin, ListLoader):
+        pass
+
+    loader = AsyncListLoader(items=list(range(100)))
+    assert list(loader) == list(range(100))
+    # second epoch works after the thread finished
+    loader2 = ListLoader(list(range(5)))
+    assert list(loader2) == [0, 1, 2, 3, 4]
