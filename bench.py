@@ -38,6 +38,10 @@ def parse_args():
                    help="disable backward/allreduce overlap (ablation)")
     p.add_argument("--fused-sgd", action="store_true",
                    help="use the CDNA4 fused SGD step kernel")
+    p.add_argument("--hipgraph", action="store_true",
+                   help="capture forward+backward in a hipGraph; allreduce + "
+                        "optimizer run after replay (trades overlap for "
+                        "launch overhead)")
     return p.parse_args()
 
 
@@ -80,23 +84,69 @@ def main():
     else:
         opt = torch.optim.SGD(model.parameters(), lr=0.0125 * hvd.size(),
                               momentum=0.9, weight_decay=5e-5)
-    opt = hvd.DistributedOptimizer(
-        opt, named_parameters=model.named_parameters(),
-        compression=compression,
-        op=hvd.Adasum if args.use_adasum else hvd.Average)
-    hvd.broadcast_parameters(model.state_dict(), root_rank=0)
-    hvd.broadcast_optimizer_state(opt, root_rank=0)
-
     loss_fn = torch.nn.CrossEntropyLoss()
     use_bf16 = cuda and not args.no_bf16
 
-    def step():
-        opt.zero_grad()
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
-            out = model(data)
-            loss = loss_fn(out, target)
-        loss.backward()
-        opt.step()
+    if args.hipgraph and cuda:
+        # hipGraph mode: fwd+bwd captured once and replayed (kills per-kernel
+        # launch overhead on the 2000-dispatch ResNet step); the gradient
+        # allreduce runs after replay as ONE grouped op, then the (fused) SGD
+        # step.  Trades backward/comm overlap for CPU-side launch cost.
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        params = [p for p in model.parameters() if p.requires_grad]
+        for p in params:
+            p.grad = torch.zeros_like(p)
+        wire = compression.wire_dtype(torch.float32)
+
+        def fwd_bwd():
+            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+                loss = loss_fn(model(data), target)
+            loss.backward()
+            return loss
+
+        # warm up allocator state on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                for p in params:
+                    p.grad.zero_()
+                fwd_bwd()
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        for p in params:
+            p.grad.zero_()
+        with torch.cuda.graph(graph):
+            fwd_bwd()
+        grads = [p.grad for p in params]
+
+        from horovod_amd.torch import mpi_ops as _ops
+
+        def step():
+            graph.replay()
+            if hvd.size() > 1 or True:
+                h = _ops._grouped_allreduce_impl(
+                    grads, grads, None, "hipgraph_grads", hvd.Average, 1.0,
+                    1.0, hvd.global_process_set, wire_dtype=wire)
+                hvd.synchronize(h)
+            opt.step()
+            for g in grads:
+                g.zero_()
+    else:
+        opt = hvd.DistributedOptimizer(
+            opt, named_parameters=model.named_parameters(),
+            compression=compression,
+            op=hvd.Adasum if args.use_adasum else hvd.Average)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        hvd.broadcast_optimizer_state(opt, root_rank=0)
+
+        def step():
+            opt.zero_grad()
+            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+                out = model(data)
+                loss = loss_fn(out, target)
+            loss.backward()
+            opt.step()
 
     for _ in range(args.warmup):
         step()
@@ -141,6 +191,7 @@ def main():
                 "compression": args.compression,
                 "reduction": "adasum" if args.use_adasum else "average",
                 "fused_sgd": args.fused_sgd,
+                "hipgraph": args.hipgraph,
             },
         }
         print(json.dumps(result), flush=True)

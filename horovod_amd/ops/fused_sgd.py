@@ -11,7 +11,7 @@ from horovod_amd import _core
 
 
 class FusedSGD(torch.optim.Optimizer):
-    def __init__(self, params, lr, momentum=0.0, weight_decay=0.0,
+    def __init__(self, params, lr=1e-3, momentum=0.0, weight_decay=0.0,
                  dampening=0.0, nesterov=False):
         if nesterov and (momentum <= 0 or dampening != 0):
             raise ValueError("nesterov requires momentum > 0, dampening 0")
