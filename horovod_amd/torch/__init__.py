@@ -32,4 +32,6 @@ from horovod_amd.common.process_sets import (ProcessSet,  # noqa: F401
 from horovod_amd.torch.optimizer import DistributedOptimizer  # noqa: F401
 from horovod_amd.torch.sync_batch_norm import SyncBatchNorm  # noqa: F401
 
+from horovod_amd.torch import elastic  # noqa: F401  (hvd.elastic.*)
+
 __version__ = "0.1.0"

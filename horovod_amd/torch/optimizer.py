@@ -233,6 +233,69 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         return super(self.__class__, self).zero_grad(*args, **kwargs)
 
 
+class _DistributedAdasumOptimizer(torch.optim.Optimizer):
+    """Adasum delta optimization (reference: optimizer.py:345-515): apply
+    the wrapped optimizer LOCALLY, Adasum-reduce the resulting weight DELTA,
+    then set w = w_start + reduced_delta.  This is the semantically correct
+    way to combine Adasum with stateful optimizers (Adam etc.)."""
+
+    def __init__(self, params, named_parameters, compression,
+                 backward_passes_per_step=1,
+                 process_set=global_process_set):
+        super(self.__class__, self).__init__(params)
+        self._compression = compression or Compression.none
+        self.backward_passes_per_step = backward_passes_per_step
+        self.process_set = process_set
+        if named_parameters is not None:
+            named = {v: k for k, v in named_parameters}
+        else:
+            named = {}
+        self._parameter_names = {}
+        for group in self.param_groups:
+            for p in group["params"]:
+                self._parameter_names[p] = named.get(
+                    p, f"adasum.noname.{len(self._parameter_names)}")
+        self._step_count = 0
+
+    def step(self, closure=None):
+        self._step_count += 1
+        if self._step_count % self.backward_passes_per_step != 0:
+            return None
+        params = [p for group in self.param_groups for p in group["params"]
+                  if p.grad is not None]
+        starts = [p.detach().clone() for p in params]
+        loss = super(self.__class__, self).step(closure)
+        if size() == 1 and not _force_allreduce():
+            return loss
+        deltas = [p.detach() - s for p, s in zip(params, starts)]
+        wire = self._compression.wire_dtype(deltas[0].dtype) if deltas else None
+        names = [f"adasum.delta.{self._parameter_names[p]}" for p in params]
+        handle = mpi_ops._allreduce_async_impl(
+            deltas, deltas, names, Adasum, 1.0, 1.0,
+            self.process_set.process_set_id,
+            mpi_ops._dtype_code(wire) if wire else
+            mpi_ops._dtype_code(deltas[0].dtype) if deltas else 5,
+            kind="grouped")
+        synchronize(handle)
+        with torch.no_grad():
+            for p, s, d in zip(params, starts, deltas):
+                p.detach().copy_(s.add_(d))
+        return loss
+
+    def synchronize(self):
+        pass
+
+    @contextmanager
+    def skip_synchronize(self):
+        yield
+
+    def zero_grad(self, *args, **kwargs):
+        return super(self.__class__, self).zero_grad(*args, **kwargs)
+
+    def reset_distributed_state(self):
+        pass
+
+
 def _force_allreduce():
     import os
     return os.environ.get("HOROVOD_ELASTIC", "0") == "1"
@@ -256,6 +319,14 @@ def DistributedOptimizer(optimizer, named_parameters=None,
             groups = num_groups
     if groups is not None and not isinstance(groups, (list, int)):
         raise ValueError("groups should be a list or int")
+
+    if op == Adasum:
+        # delta optimization (reference: factory optimizer.py:516+ routes
+        # op=Adasum to _DistributedAdasumOptimizer)
+        cls = type(optimizer.__class__.__name__, (optimizer.__class__,),
+                   dict(_DistributedAdasumOptimizer.__dict__))
+        return cls(optimizer.param_groups, named_parameters, compression,
+                   backward_passes_per_step, process_set)
 
     cls = type(optimizer.__class__.__name__, (optimizer.__class__,),
                dict(_DistributedOptimizer.__dict__))

@@ -90,3 +90,13 @@ def test_async_data_loader():
     # second epoch works after the thread finished
     loader2 = ListLoader(list(range(5)))
     assert list(loader2) == [0, 1, 2, 3, 4]
+
+
+def test_spark_ray_graceful_without_deps():
+    import horovod_amd.spark as hvd_spark
+    import horovod_amd.ray as hvd_ray
+    with pytest.raises(ImportError):
+        hvd_spark.run(lambda: 1)
+    ex = hvd_ray.RayExecutor(num_workers=2)
+    with pytest.raises(ImportError):
+        ex.start()

@@ -272,3 +272,26 @@ def test_autotune_np2(tmp_path):
         # mechanism is alive: threshold is a sane positive value
         assert core.get_fusion_threshold() > 0
     """, extra_env={"HOROVOD_AUTOTUNE": "1", "HOROVOD_AUTOTUNE_LOG": log})
+
+
+def test_adasum_optimizer_np2():
+    """op=Adasum routes to the delta optimizer; params stay identical across
+    ranks (reference: _DistributedAdasumOptimizer)."""
+    run_workers(2, """
+        torch.manual_seed(21)
+        model = torch.nn.Linear(6, 2)
+        opt = hvd.DistributedOptimizer(
+            torch.optim.Adam(model.parameters(), lr=0.01),
+            named_parameters=model.named_parameters(), op=hvd.Adasum)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(77 + rank)
+        for step in range(4):
+            opt.zero_grad()
+            loss = model(torch.randn(5, 6)).sum()
+            loss.backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        g = hvd.allgather(flat.unsqueeze(0), name="ada_opt")
+        assert torch.allclose(g[0], g[1], atol=1e-6), \
+            (g[0] - g[1]).abs().max()
+    """)
