@@ -27,7 +27,10 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch-size", type=int, default=64, help="per-GPU batch")
-    p.add_argument("--model", default="resnet50")
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "resnet101", "resnet152",
+                            "bert-large", "bert-base"])
+    p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--compression", default="none",
                    choices=["none", "fp16", "bf16"])
@@ -58,20 +61,30 @@ def main():
     else:
         device = torch.device("cpu")
 
-    from horovod_amd.models import resnet50, resnet101, resnet152
+    from horovod_amd.models import (bert_base, bert_large, resnet50,
+                                    resnet101, resnet152)
+    is_bert = args.model.startswith("bert")
     model_fn = {"resnet50": resnet50, "resnet101": resnet101,
-                "resnet152": resnet152}[args.model]
+                "resnet152": resnet152, "bert-large": bert_large,
+                "bert-base": bert_base}[args.model]
     torch.manual_seed(42)
     model = model_fn().to(device)
-    if cuda:
+    if cuda and not is_bert:
         model = model.to(memory_format=torch.channels_last)
 
-    batch = args.batch_size if cuda else 8
-    image_size = args.image_size if cuda else 64
-    data = torch.randn(batch, 3, image_size, image_size, device=device)
-    if cuda:
-        data = data.to(memory_format=torch.channels_last)
-    target = torch.randint(0, 1000, (batch,), device=device)
+    if is_bert:
+        batch = args.batch_size if cuda else 2
+        seq = args.seq_len if cuda else 32
+        image_size = None
+        data = torch.randint(0, 30522, (batch, seq), device=device)
+        target = torch.randint(0, 30522, (batch, seq), device=device)
+    else:
+        batch = args.batch_size if cuda else 8
+        image_size = args.image_size if cuda else 64
+        data = torch.randn(batch, 3, image_size, image_size, device=device)
+        if cuda:
+            data = data.to(memory_format=torch.channels_last)
+        target = torch.randint(0, 1000, (batch,), device=device)
 
     from horovod_amd.torch.compression import Compression
     compression = {"none": Compression.none, "fp16": Compression.fp16,
@@ -143,8 +156,11 @@ def main():
         def step():
             opt.zero_grad()
             with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
-                out = model(data)
-                loss = loss_fn(out, target)
+                if is_bert:
+                    mlm, _ = model(data)
+                    loss = loss_fn(mlm.flatten(0, 1), target.flatten())
+                else:
+                    loss = loss_fn(model(data), target)
             loss.backward()
             opt.step()
 
@@ -171,9 +187,10 @@ def main():
     value = total_images / elapsed
     if hvd.rank() == 0:
         result = {
-            "metric": "images/sec, ResNet-50 synthetic (total over all GPUs)",
+            "metric": (f"sequences/sec, {args.model} synthetic (total)" if is_bert
+                       else "images/sec, ResNet-50 synthetic (total over all GPUs)"),
             "value": round(value, 2),
-            "unit": "images/sec",
+            "unit": "sequences/sec" if is_bert else "images/sec",
             "n_gpus": n,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -187,6 +204,7 @@ def main():
                 "model": args.model,
                 "global_batch": n * batch,
                 "image_size": image_size,
+                "seq_len": args.seq_len if is_bert else None,
                 "parallelism": f"dp{n}",
                 "compression": args.compression,
                 "reduction": "adasum" if args.use_adasum else "average",
