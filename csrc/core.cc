@@ -198,7 +198,8 @@ void UnpackInto(TensorTableEntry& e, at::Tensor flat_slice) {
   at::Tensor src = flat_slice;
   if (e.postscale != 1.0) src = src.to(at::kDouble).mul_(e.postscale);
   if (!e.output.defined()) e.output = at::empty_like(e.tensor);
-  e.output.flatten().copy_(src.reshape(e.output.sizes()).flatten());
+  // copy_ handles layout + dtype (output may be strided)
+  e.output.copy_(src.reshape(e.output.sizes()));
 }
 
 // Adasum pairwise combine: a = a*(1 - dot/(2|a|^2)) + b*(1 - dot/(2|b|^2)),

@@ -175,8 +175,7 @@ def _allreduce_async_impl(tensors, outputs, name, true_op, pre, post, ps_id,
 def allreduce_async(tensor, average=None, name=None, op=None,
                     prescale_factor=1.0, postscale_factor=1.0,
                     process_set=global_process_set):
-    output = torch.empty_like(tensor)
-    return _do_allreduce_async(tensor, output, average, name, op,
+    return _do_allreduce_async(tensor, None, average, name, op,
                                prescale_factor, postscale_factor, process_set)
 
 
@@ -189,10 +188,14 @@ def allreduce_async_(tensor, average=None, name=None, op=None,
 
 def _do_allreduce_async(tensor, output, average, name, op, prescale_factor,
                         postscale_factor, process_set, wire_dtype=None):
-    if tensor.dtype == torch.bool:
-        # bool allreduce == logical or via max (reference treats bool via
-        # custom MPI op; RCCL has no bool type)
-        pass
+    if not tensor.is_contiguous():
+        if output is tensor:
+            raise ValueError(
+                "hvd.allreduce_ requires a contiguous tensor; call "
+                ".contiguous() first or use the out-of-place hvd.allreduce")
+        tensor = tensor.contiguous()
+    if output is None:
+        output = torch.empty_like(tensor)
     true_op, _, pre, post = _resolve_scales(op, average, prescale_factor,
                                             postscale_factor, process_set)
     ps_id = _set_id(process_set)
@@ -276,6 +279,14 @@ def grouped_allreduce_async_(tensors, average=None, name=None, op=None,
 def _grouped_allreduce_impl(tensors, outputs, average, name, op,
                             prescale_factor, postscale_factor, process_set,
                             wire_dtype=None):
+    tensors = list(tensors)
+    outputs = list(outputs)
+    for i, t in enumerate(tensors):
+        if not t.is_contiguous():
+            if outputs[i] is t:
+                raise ValueError(
+                    "grouped_allreduce_ requires contiguous tensors")
+            tensors[i] = t.contiguous()
     true_op, _, pre, post = _resolve_scales(op, average, prescale_factor,
                                             postscale_factor, process_set)
     ps_id = _set_id(process_set)

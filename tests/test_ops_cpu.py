@@ -242,3 +242,17 @@ def test_parallel_grid_sets_np4():
         expected = (col + 1) + (col + 3)
         assert out[0].item() == expected, (out, expected)
     """)
+
+
+def test_noncontiguous_np2():
+    run_workers(2, """
+        base = torch.arange(12).float().reshape(3, 4)
+        t = base.t()  # non-contiguous view
+        out = hvd.allreduce(t, average=False, name="nc")
+        assert torch.allclose(out, base.t() * 2), out
+        try:
+            hvd.allreduce_(t, average=False, name="nc2")
+            raise SystemExit("expected error for in-place non-contiguous")
+        except ValueError:
+            pass
+    """)
