@@ -422,11 +422,16 @@ void Execute(GlobalState& st, Response& resp,
   switch (resp.type) {
     case ResponseType::ALLREDUCE: {
       activity = "RCCL_ALLREDUCE";
-      bool direct = entries.size() == 1 && entries[0].prescale == 1.0 &&
-                    entries[0].postscale == 1.0 &&
-                    DataTypeFromTorch(entries[0].tensor.scalar_type()) == wire &&
-                    entries[0].tensor.is_contiguous() &&
-                    entries[0].output.defined() && entries[0].output.is_contiguous();
+      // dense (possibly permuted, e.g. channels_last) tensors are raw-
+      // copyable: identical layout on every rank, so RCCL/pack operate in
+      // memory order consistently.
+      bool direct =
+          entries.size() == 1 && entries[0].prescale == 1.0 &&
+          entries[0].postscale == 1.0 &&
+          DataTypeFromTorch(entries[0].tensor.scalar_type()) == wire &&
+          entries[0].tensor.is_non_overlapping_and_dense() &&
+          entries[0].output.defined() &&
+          entries[0].output.strides() == entries[0].tensor.strides();
       if (direct) {
         auto& e = entries[0];
         RCCL_CHECK(ncclAllReduce(e.tensor.data_ptr(), e.output.data_ptr(),
@@ -455,8 +460,10 @@ void Execute(GlobalState& st, Response& resp,
       activity = "RCCL_BCAST";
       auto& e = entries[0];
       int root_li = set.local_index(resp.root_rank);
-      at::Tensor in = e.tensor.is_contiguous() ? e.tensor : e.tensor.contiguous();
-      if (!e.output.defined()) e.output = e.tensor;
+      at::Tensor in = e.tensor.is_non_overlapping_and_dense()
+                          ? e.tensor
+                          : e.tensor.contiguous();
+      if (!e.output.defined()) e.output = in;
       RCCL_CHECK(ncclBroadcast(in.data_ptr(), e.output.data_ptr(), in.numel(),
                                wire_nccl, root_li, comm, stream));
       break;

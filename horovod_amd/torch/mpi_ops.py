@@ -79,6 +79,20 @@ def _dtype_code(dtype):
         raise ValueError(f"horovod_amd: unsupported dtype {dtype}")
 
 
+def _dense_ok(t):
+    """True when the tensor occupies a dense (non-overlapping) block of
+    memory, possibly with permuted strides (e.g. channels_last).  Dense
+    tensors are raw-copyable: every rank shares the identical layout, so the
+    pack kernels and RCCL operate in memory order consistently."""
+    if t.is_contiguous():
+        return True
+    if t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last):
+        return True
+    if t.dim() == 5 and t.is_contiguous(memory_format=torch.channels_last_3d):
+        return True
+    return False
+
+
 def _next_name(prefix):
     global _NULL_NAME_COUNTER
     _NULL_NAME_COUNTER += 1
@@ -188,11 +202,11 @@ def allreduce_async_(tensor, average=None, name=None, op=None,
 
 def _do_allreduce_async(tensor, output, average, name, op, prescale_factor,
                         postscale_factor, process_set, wire_dtype=None):
-    if not tensor.is_contiguous():
+    if not _dense_ok(tensor):
         if output is tensor:
             raise ValueError(
-                "hvd.allreduce_ requires a contiguous tensor; call "
-                ".contiguous() first or use the out-of-place hvd.allreduce")
+                "hvd.allreduce_ requires a dense tensor; call .contiguous() "
+                "first or use the out-of-place hvd.allreduce")
         tensor = tensor.contiguous()
     if output is None:
         output = torch.empty_like(tensor)
@@ -282,10 +296,9 @@ def _grouped_allreduce_impl(tensors, outputs, average, name, op,
     tensors = list(tensors)
     outputs = list(outputs)
     for i, t in enumerate(tensors):
-        if not t.is_contiguous():
+        if not _dense_ok(t):
             if outputs[i] is t:
-                raise ValueError(
-                    "grouped_allreduce_ requires contiguous tensors")
+                raise ValueError("grouped_allreduce_ requires dense tensors")
             tensors[i] = t.contiguous()
     true_op, _, pre, post = _resolve_scales(op, average, prescale_factor,
                                             postscale_factor, process_set)
@@ -393,6 +406,8 @@ def grouped_allgather(tensors, name=None, process_set=global_process_set):
 # ---------------------------------------------------------------------------
 def broadcast_async(tensor, root_rank, name=None,
                     process_set=global_process_set):
+    if not _dense_ok(tensor):
+        tensor = tensor.contiguous()
     output = torch.empty_like(tensor)
     name = name or _next_name("broadcast")
     h = _core.broadcast_async(tensor, output, root_rank, "broadcast." + name,
@@ -402,6 +417,8 @@ def broadcast_async(tensor, root_rank, name=None,
 
 def broadcast_async_(tensor, root_rank, name=None,
                      process_set=global_process_set):
+    if not _dense_ok(tensor):
+        raise ValueError("hvd.broadcast_ requires a dense tensor")
     name = name or _next_name("broadcast")
     h = _core.broadcast_async(tensor, tensor, root_rank, "broadcast." + name,
                               _set_id(process_set))

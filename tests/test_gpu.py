@@ -164,3 +164,19 @@ def test_fused_sgd_matches_torch(hvd):
     for rp, fp in zip(ref_params, fused_params):
         assert torch.allclose(rp, fp, rtol=1e-5, atol=1e-6), \
             (rp - fp).abs().max()
+
+
+@requires_gpu
+def test_channels_last_dense_layout(hvd):
+    """Dense-but-permuted (channels_last) tensors: raw memory-order
+    collectives must round-trip exactly (the layout every ResNet conv
+    grad/param has under channels_last training)."""
+    t = torch.randn(8, 3, 4, 4, device="cuda").to(
+        memory_format=torch.channels_last)
+    out = hvd.allreduce(t, average=False, name="cl1")
+    assert torch.equal(out, t)
+    t2 = t.clone()
+    hvd.broadcast_(t2, root_rank=0, name="cl2")
+    assert torch.equal(t2, t)
+    outs = hvd.grouped_allreduce([t, t.clone()], average=False, name="cl3")
+    assert torch.equal(outs[0], t) and torch.equal(outs[1], t)
