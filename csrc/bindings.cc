@@ -45,12 +45,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         [](int rank, int size, int local_rank, int local_size, int cross_rank,
            int cross_size, const std::string& addr, int port,
            int64_t fusion_threshold, double cycle_time_ms, int cache_capacity,
-           double stall_warning_sec, bool timeline) {
+           double stall_warning_sec, double stall_shutdown_sec, bool timeline) {
           ControllerConfig cfg;
           cfg.fusion_threshold_bytes = fusion_threshold;
           cfg.cycle_time_ms = cycle_time_ms;
           cfg.cache_capacity = (size_t)cache_capacity;
           cfg.stall_warning_sec = stall_warning_sec;
+          cfg.stall_shutdown_sec = stall_shutdown_sec;
           cfg.timeline_enabled = timeline;
           py::gil_scoped_release nogil;
           InitHorovod(rank, size, local_rank, local_size, cross_rank, cross_size,
@@ -60,7 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("local_size"), py::arg("cross_rank"), py::arg("cross_size"),
         py::arg("addr"), py::arg("port"), py::arg("fusion_threshold"),
         py::arg("cycle_time_ms"), py::arg("cache_capacity"),
-        py::arg("stall_warning_sec"), py::arg("timeline"));
+        py::arg("stall_warning_sec"), py::arg("stall_shutdown_sec"),
+        py::arg("timeline"));
 
   m.def("shutdown", [] {
     py::gil_scoped_release nogil;

@@ -27,7 +27,21 @@ std::string SetKey(int32_t set_id, const std::string& name) {
 bool SameSignature(const Request& a, const Request& b) {
   return a.type == b.type && a.dtype == b.dtype && a.shape == b.shape &&
          a.root_rank == b.root_rank && a.reduce_op == b.reduce_op &&
-         a.process_set_id == b.process_set_id && a.device == b.device;
+         a.process_set_id == b.process_set_id && a.device == b.device &&
+         a.group_key == b.group_key && a.group_size == b.group_size;
+}
+
+// merge single-tensor responses of one group into one fused response
+// (bypasses the fusion threshold — reference GroupTable semantics)
+Response MergeGroup(std::vector<Response>& singles) {
+  Response out = singles.front();
+  for (size_t i = 1; i < singles.size(); ++i) {
+    out.names.push_back(singles[i].names[0]);
+    out.tensor_shapes.insert(out.tensor_shapes.end(),
+                             singles[i].tensor_shapes.begin(),
+                             singles[i].tensor_shapes.end());
+  }
+  return out;
 }
 
 bool IsFusableType(ResponseType t) {
@@ -107,6 +121,8 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
     single.process_set_id = response.process_set_id;
     single.device = response.device;
     single.root_rank = response.root_rank;
+    single.group_key = response.group_key;
+    single.group_size = response.group_size;
     if (nsizes_per)
       single.tensor_sizes.assign(response.tensor_sizes.begin() + i * nsizes_per,
                                  response.tensor_sizes.begin() + (i + 1) * nsizes_per);
@@ -129,6 +145,8 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
     if (i < reqs.size()) sig.reduce_op = reqs[i].reduce_op;
     sig.process_set_id = response.process_set_id;
     sig.device = response.device;
+    sig.group_key = response.group_key;
+    sig.group_size = response.group_size;
 
     std::string key = SetKey(sig.process_set_id, sig.name);
     auto it = name_to_slot_.find(key);
@@ -261,20 +279,41 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     // ran, the OR flag may be false on every rank.  Both evictor and owner
     // see the same OR bits, so every rank whose pending got displaced flags
     // the next cycle via non-empty `slow`/`inflight_` — we just hold them.
-    std::deque<Response> fast;
+    // collect common-hit slots; grouped tensors only fire when the whole
+    // group is commonly ready (strict GroupTable semantics)
+    std::vector<int> hit_slots;
+    std::unordered_map<std::string, std::vector<int>> group_slots;
     for (size_t s = 0; s < cache_.num_slots(); ++s) {
       if (!(vecA[s / 8] & (1 << (s % 8)))) continue;
       if (vecB[s / 8] & (1 << (s % 8))) continue;  // invalidated this cycle
       if (!cache_.slot_live((int)s)) continue;
-      fast.push_back(cache_.Get((int)s));
-      // pop from my pending (members only have it)
+      const Response& r = cache_.Get((int)s);
+      if (!r.group_key.empty() && r.group_size > 1)
+        group_slots[SetKey(r.process_set_id, r.group_key)].push_back((int)s);
+      else
+        hit_slots.push_back((int)s);
+    }
+    std::deque<Response> fast;
+    auto consume = [&](int s) {
+      const Response& r = cache_.Get(s);
       for (auto it = cached_pending_.begin(); it != cached_pending_.end(); ++it) {
         if (SetKey(it->process_set_id, it->name) ==
-            SetKey(cache_.Get((int)s).process_set_id, cache_.Get((int)s).names[0])) {
+            SetKey(r.process_set_id, r.names[0])) {
           cached_pending_.erase(it);
           break;
         }
       }
+      return r;
+    };
+    for (int s : hit_slots) fast.push_back(consume(s));
+    for (auto& kv : group_slots) {
+      auto& slots = kv.second;
+      if ((int32_t)slots.size() != cache_.Get(slots[0]).group_size)
+        continue;  // incomplete group: stays pending, re-votes next cycle
+      std::sort(slots.begin(), slots.end());
+      std::vector<Response> singles;
+      for (int s : slots) singles.push_back(consume(s));
+      fast.push_back(MergeGroup(singles));
     }
     auto fast_fused = FuseResponses(fast);
     for (auto& r : fast_fused) result.responses.push_back(std::move(r));
@@ -409,7 +448,21 @@ std::vector<Response> Controller::CoordinatorProcess(
       needed -= joined_needed;
     }
     if (pt.count >= needed) {
-      ready_responses_.push_back(ConstructResponse(it->first, pt));
+      Response r = ConstructResponse(it->first, pt);
+      if (!r.group_key.empty() && r.group_size > 1 &&
+          r.type != ResponseType::ERROR) {
+        // strict group semantics (reference GroupTable): hold until every
+        // member of the group is ready, then emit ONE fused response.
+        auto& hold = group_hold_[SetKey(r.process_set_id, r.group_key)];
+        hold.push_back(std::move(r));
+        if ((int32_t)hold.size() == hold.front().group_size) {
+          ready_responses_.push_back(MergeGroup(hold));
+          group_hold_.erase(SetKey(hold.front().process_set_id,
+                                   hold.front().group_key));
+        }
+      } else {
+        ready_responses_.push_back(std::move(r));
+      }
       table_.erase(it);
       ko = arrival_order.erase(ko);
     } else {
@@ -455,6 +508,8 @@ Response Controller::ConstructResponse(const std::string& key, PendingTensor& pt
   resp.process_set_id = first.process_set_id;
   resp.device = first.device;
   resp.root_rank = first.root_rank;
+  resp.group_key = first.group_key;
+  resp.group_size = first.group_size;
 
   // Validate cross-rank consistency (reference: controller.cc
   // ConstructResponse:496-843 error text behavior).
@@ -623,6 +678,12 @@ void Controller::CheckForStalledTensors() {
                  "[horovod_amd] WARNING: tensor %s stalled for %.0fs; waiting on "
                  "ranks: %s\n",
                  kv.first.c_str(), age, missing.str().c_str());
+    if (cfg_.stall_shutdown_sec > 0 && age > cfg_.stall_shutdown_sec) {
+      // reference: HOROVOD_STALL_SHUTDOWN_TIME_SECONDS aborts the job when a
+      // stall persists (stall_inspector.h:30-97)
+      throw std::runtime_error("stalled tensor " + kv.first + " exceeded "
+                               "HOROVOD_STALL_SHUTDOWN_TIME_SECONDS; aborting");
+    }
   }
 }
 

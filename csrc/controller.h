@@ -162,6 +162,8 @@ class Controller {
   std::unordered_map<std::string, PendingTensor> table_;
   std::deque<std::string> arrival_order_;  // table keys, first-seen order
   std::deque<Response> ready_responses_;  // completed, awaiting fusion window
+  // grouped-op holding area: set:group_key -> completed singles
+  std::unordered_map<std::string, std::vector<Response>> group_hold_;
   bool join_in_progress_ = false;
 
   std::unordered_map<int32_t, ProcessSetInfo> process_sets_;

@@ -1,5 +1,8 @@
 #include "core.h"
 
+#include <pthread.h>
+#include <sched.h>
+
 #include <chrono>
 #include <cstdio>
 #include <cstring>
@@ -585,7 +588,19 @@ void Abort(GlobalState& st, const std::string& why) {
   st.handles.FailAll(s);
 }
 
+void SetThreadAffinity() {
+  // reference: parse_and_set_affinity (common.cc) / HOROVOD_THREAD_AFFINITY
+  const char* aff = std::getenv("HOROVOD_THREAD_AFFINITY");
+  if (!aff) return;
+  int cpu = atoi(aff) + State().local_rank;
+  cpu_set_t set;
+  CPU_ZERO(&set);
+  CPU_SET(cpu % CPU_SETSIZE, &set);
+  pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
+}
+
 void BackgroundLoop(GlobalState& st) {
+  SetThreadAffinity();
   // Shutdown protocol: a graceful exit needs every rank's agreement (AND bit
   // in the cache-sync round) so no peer is left mid-collective; but a rank
   // must never hang forever on a dead peer — after the grace window it
@@ -753,8 +768,15 @@ int EnqueueAllreduceMulti(std::vector<at::Tensor> tensors,
   std::vector<Request> reqs;
   std::vector<TensorTableEntry> entries;
   for (size_t i = 0; i < tensors.size(); ++i) {
-    reqs.push_back(MakeRequest(RequestType::ALLREDUCE, names[i], tensors[i], op,
-                               prescale, postscale, -1, process_set_id, wire_dtype));
+    auto req = MakeRequest(RequestType::ALLREDUCE, names[i], tensors[i], op,
+                           prescale, postscale, -1, process_set_id, wire_dtype);
+    if (tensors.size() > 1) {
+      // strict group identity (reference: GroupTable) — the whole group
+      // negotiates and executes as one fused unit
+      req.group_key = names[0];
+      req.group_size = (int32_t)tensors.size();
+    }
+    reqs.push_back(std::move(req));
     entries.push_back(MakeEntry(names[i], tensors[i], outputs[i], op, prescale,
                                 postscale, -1, process_set_id));
   }

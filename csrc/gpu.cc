@@ -2,6 +2,7 @@
 
 #include <hip/hip_runtime_api.h>
 #include <rccl/rccl.h>
+#include <roctracer/roctx.h>
 
 #include <c10/hip/HIPCachingAllocator.h>
 #include <c10/hip/HIPGuard.h>
@@ -418,6 +419,13 @@ void Execute(GlobalState& st, Response& resp,
   std::vector<hipEvent_t> ready;
   WaitReadyEvents(ctx, entries, ready);
 
+  // roctx range so external profilers (rocprofv3 --marker-trace) see the op
+  // (reference: SharedNvtxOpRange, nvtx_op_range.{cc,h}); disable via
+  // HOROVOD_DISABLE_ROCTX_RANGES.
+  static const bool roctx_on = std::getenv("HOROVOD_DISABLE_ROCTX_RANGES") == nullptr;
+  if (roctx_on)
+    roctxRangePushA((std::string("hvd.") + entries[0].name).c_str());
+
   const char* activity = "RCCL_OP";
   switch (resp.type) {
     case ResponseType::ALLREDUCE: {
@@ -602,6 +610,7 @@ void Execute(GlobalState& st, Response& resp,
       throw std::runtime_error("horovod_amd: bad GPU response type");
   }
 
+  if (roctx_on) roctxRangePop();
   Finalize(ctx, std::move(entries), std::move(ready), activity, t_start);
 }
 

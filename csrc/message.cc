@@ -136,6 +136,8 @@ void Request::Serialize(std::string& out) const {
   put<int32_t>(out, process_set_id);
   put<int32_t>(out, device);
   put_i64vec(out, splits);
+  put_str(out, group_key);
+  put<int32_t>(out, group_size);
 }
 
 Request Request::Deserialize(const char*& p, const char* end) {
@@ -152,6 +154,8 @@ Request Request::Deserialize(const char*& p, const char* end) {
   r.process_set_id = get<int32_t>(p, end);
   r.device = get<int32_t>(p, end);
   r.splits = get_i64vec(p, end);
+  r.group_key = get_str(p, end);
+  r.group_size = get<int32_t>(p, end);
   return r;
 }
 
@@ -185,6 +189,8 @@ void Response::Serialize(std::string& out) const {
   put<int32_t>(out, root_rank);
   put<int32_t>(out, last_joined_rank);
   put_i64vec(out, tensor_shapes);
+  put_str(out, group_key);
+  put<int32_t>(out, group_size);
 }
 
 Response Response::Deserialize(const char*& p, const char* end) {
@@ -202,6 +208,8 @@ Response Response::Deserialize(const char*& p, const char* end) {
   r.root_rank = get<int32_t>(p, end);
   r.last_joined_rank = get<int32_t>(p, end);
   r.tensor_shapes = get_i64vec(p, end);
+  r.group_key = get_str(p, end);
+  r.group_size = get<int32_t>(p, end);
   return r;
 }
 

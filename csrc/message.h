@@ -53,6 +53,10 @@ struct Request {
   int32_t process_set_id = 0;
   int32_t device = CPU_DEVICE_ID;
   std::vector<int64_t> splits;  // alltoall send splits (set-local order)
+  // grouped ops (reference: GroupTable): all tensors sharing group_key must
+  // negotiate and execute as ONE fused unit of group_size tensors.
+  std::string group_key;
+  int32_t group_size = 0;
 
   void Serialize(std::string& out) const;
   static Request Deserialize(const char*& p, const char* end);
@@ -85,6 +89,9 @@ struct Response {
   // Shapes of the tensors (flattened [ndims..., -1 terminator] per name) so a
   // joined rank can allocate zero substitutes.  Populated for ALLREDUCE.
   std::vector<int64_t> tensor_shapes;
+  // grouped-op identity (see Request)
+  std::string group_key;
+  int32_t group_size = 0;
 
   void Serialize(std::string& out) const;
   static Response Deserialize(const char*& p, const char* end);

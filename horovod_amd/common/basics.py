@@ -60,6 +60,8 @@ class HorovodBasics:
         cycle_ms = float(os.environ.get("HOROVOD_CYCLE_TIME", 1.0))
         cache_cap = _env_int(["HOROVOD_CACHE_CAPACITY"], 1024)
         stall_sec = float(os.environ.get("HOROVOD_STALL_CHECK_TIME_SECONDS", 60.0))
+        stall_shutdown = float(os.environ.get(
+            "HOROVOD_STALL_SHUTDOWN_TIME_SECONDS", 0.0))
         timeline = bool(os.environ.get("HOROVOD_TIMELINE"))
 
         _core.init(rank=rank, size=size, local_rank=local_rank,
@@ -67,7 +69,7 @@ class HorovodBasics:
                    cross_size=cross_size, addr=addr, port=port,
                    fusion_threshold=int(fusion_mb), cycle_time_ms=cycle_ms,
                    cache_capacity=cache_cap, stall_warning_sec=stall_sec,
-                   timeline=timeline)
+                   stall_shutdown_sec=stall_shutdown, timeline=timeline)
         self._initialized_here = True
         atexit.register(self.shutdown)
 

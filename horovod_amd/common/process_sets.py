@@ -62,6 +62,11 @@ def add_process_set(process_set):
     if process_set.process_set_id is not None:
         raise ValueError("process set already registered")
     process_set.process_set_id = _core.add_process_set(process_set.ranks)
+    # registration is collective (must be called identically on every rank,
+    # reference: horovod_add_process_set operations.cc:1262-1328); barrier so
+    # no rank races ahead and uses the set before peers registered it
+    h = _core.barrier_async(0)
+    _core.wait(h)
     return process_set
 
 
