@@ -134,6 +134,9 @@ class Controller {
   void set_cycle_time_ms(double ms) { cfg_.cycle_time_ms = ms; }
   double cycle_time_ms() const { return cfg_.cycle_time_ms; }
   ControllerConfig& config() { return cfg_; }
+  // tensors still negotiating (cache-hit bits not yet agreed, or requests at
+  // the coordinator) — the cycle loop must not sleep while any exist
+  bool has_pending() const { return !cached_pending_.empty() || !inflight_.empty(); }
   void ApplyTune(int64_t fusion_bytes, double cycle_time_ms) {
     cfg_.fusion_threshold_bytes = fusion_bytes;
     cfg_.cycle_time_ms = cycle_time_ms;

@@ -33,6 +33,7 @@ Status TensorQueue::Add(Request req, TensorTableEntry entry) {
   }
   table_.emplace(std::move(key), std::move(entry));
   messages_.push_back(std::move(req));
+  cv_.notify_one();
   return Status::OK();
 }
 
@@ -48,7 +49,19 @@ Status TensorQueue::AddMulti(std::vector<Request>& reqs,
                    std::move(entries[i]));
     messages_.push_back(std::move(reqs[i]));
   }
+  cv_.notify_one();
   return Status::OK();
+}
+
+bool TensorQueue::has_messages() const {
+  std::lock_guard<std::mutex> g(mu_);
+  return !messages_.empty();
+}
+
+void TensorQueue::WaitForMessages(double ms) {
+  std::unique_lock<std::mutex> lk(mu_);
+  cv_.wait_for(lk, std::chrono::duration<double, std::milli>(ms),
+               [&] { return !messages_.empty(); });
 }
 
 std::vector<Request> TensorQueue::PopMessages() {
@@ -628,11 +641,13 @@ void BackgroundLoop(GlobalState& st) {
     }
     for (auto& resp : rl.responses) PerformOperation(st, resp);
     if (rl.shutdown) break;
-    if (rl.responses.empty()) {
-      auto elapsed = std::chrono::steady_clock::now() - cycle_start;
-      auto target =
-          std::chrono::duration<double, std::milli>(st.controller->cycle_time_ms());
-      if (elapsed < target) std::this_thread::sleep_for(target - elapsed);
+    if (rl.responses.empty() && !st.queue.has_messages() &&
+        !st.controller->has_pending()) {
+      auto elapsed = std::chrono::duration<double, std::milli>(
+                         std::chrono::steady_clock::now() - cycle_start)
+                         .count();
+      double remain = st.controller->cycle_time_ms() - elapsed;
+      if (remain > 0) st.queue.WaitForMessages(remain);
     }
   }
   gpu::WaitAllPending();

@@ -36,9 +36,14 @@ class TensorQueue {
   bool PopEntry(int32_t set_id, const std::string& name, TensorTableEntry& out);
   void FailAll(const Status& s);
   size_t size() const;
+  bool has_messages() const;
+  // Sleep up to `ms`, waking immediately when a new request is enqueued —
+  // the trailing gradient bucket of a step never waits out a cycle.
+  void WaitForMessages(double ms);
 
  private:
   mutable std::mutex mu_;
+  std::condition_variable cv_;
   std::deque<Request> messages_;
   std::unordered_map<std::string, TensorTableEntry> table_;
 };
