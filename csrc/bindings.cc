@@ -120,6 +120,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("poll", [](int handle) { return State().handles.Poll(handle); });
   m.def("wait", &WaitHandle);
 
+  m.def("adasum_combine_", [](std::vector<at::Tensor> a,
+                              std::vector<at::Tensor> b) {
+    hvd::gpu::AdasumCombine(a, b);
+  });
+
   // ---- fused optimizer kernels -------------------------------------------
   m.def("fused_sgd_step",
         [](std::vector<at::Tensor> params, std::vector<at::Tensor> grads,

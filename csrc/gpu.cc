@@ -614,6 +614,38 @@ void Execute(GlobalState& st, Response& resp,
   Finalize(ctx, std::move(entries), std::move(ready), activity, t_start);
 }
 
+void AdasumCombine(std::vector<at::Tensor>& a, std::vector<at::Tensor>& b) {
+  if (a.empty()) return;
+  int device = (int)a[0].get_device();
+  c10::hip::HIPGuard guard(device);
+  hipStream_t stream = c10::hip::getCurrentHIPStream(device).stream();
+  at::Tensor dots_t = at::zeros(
+      {(int64_t)a.size() * 3},
+      at::TensorOptions().dtype(at::kDouble).device(at::kCUDA, device));
+  double* dots = dots_t.data_ptr<double>();
+  int dt = (int)DataTypeFromTorch(a[0].scalar_type());
+  for (size_t start = 0; start < a.size(); start += kCopyBatchCapacity) {
+    AdasumBatchArgs args;
+    args.count = (int)std::min<size_t>(kCopyBatchCapacity, a.size() - start);
+    for (int k = 0; k < args.count; ++k) {
+      args.a[k] = a[start + k].data_ptr();
+      args.b[k] = b[start + k].data_ptr();
+      args.numel[k] = (unsigned long long)a[start + k].numel();
+    }
+    HIP_CHECK(AdasumDotsLaunch(args, dt, dots + start * 3, stream));
+  }
+  for (size_t start = 0; start < a.size(); start += kCopyBatchCapacity) {
+    AdasumBatchArgs args;
+    args.count = (int)std::min<size_t>(kCopyBatchCapacity, a.size() - start);
+    for (int k = 0; k < args.count; ++k) {
+      args.a[k] = a[start + k].data_ptr();
+      args.b[k] = b[start + k].data_ptr();
+      args.numel[k] = (unsigned long long)a[start + k].numel();
+    }
+    HIP_CHECK(AdasumScaledAddLaunch(args, dt, dots + start * 3, stream));
+  }
+}
+
 void FusedSgdStep(std::vector<at::Tensor>& params,
                   std::vector<at::Tensor>& grads,
                   std::vector<at::Tensor>& momenta, double lr, double momentum,

@@ -37,6 +37,11 @@ void Shutdown();
 // the native path ran).
 bool RcclUsed();
 
+// Direct Adasum pairwise combine on torch's current stream:
+// a[i] = acoef*a[i] + bcoef*b[i] per tensor (kernel unit-test surface; the
+// collective path uses the same kernels inside ExecuteAdasum).
+void AdasumCombine(std::vector<at::Tensor>& a, std::vector<at::Tensor>& b);
+
 // Fused SGD step on torch's current stream (one kernel for all buckets;
 // empty `momenta` = plain SGD).  fp32 tensors.
 void FusedSgdStep(std::vector<at::Tensor>& params,

@@ -180,3 +180,29 @@ def test_channels_last_dense_layout(hvd):
     assert torch.equal(t2, t)
     outs = hvd.grouped_allreduce([t, t.clone()], average=False, name="cl3")
     assert torch.equal(outs[0], t) and torch.equal(outs[1], t)
+
+
+@requires_gpu
+def test_adasum_kernels_golden_gpu(hvd):
+    """Drive adasum_dots_k / adasum_scaledadd_k directly and compare against
+    the fp64 torch formula (per-tensor coefficient isolation)."""
+    from horovod_amd import _core
+    torch.manual_seed(11)
+    for dtype, tol in ((torch.float32, 1e-6), (torch.bfloat16, 5e-2),
+                       (torch.float64, 1e-12)):
+        a = [torch.randn(n, device="cuda").to(dtype) for n in (17, 1024, 65537)]
+        b = [torch.randn(n, device="cuda").to(dtype) for n in (17, 1024, 65537)]
+        expected = []
+        for x, y in zip(a, b):
+            xd, yd = x.double(), y.double()
+            dot = xd.dot(yd)
+            na, nb = xd.dot(xd), yd.dot(yd)
+            ac = 1 - dot / (2 * na) if na > 0 else 1.0
+            bc = 1 - dot / (2 * nb) if nb > 0 else 1.0
+            expected.append((ac * xd + bc * yd).to(dtype))
+        _core.adasum_combine_(a, b)
+        torch.cuda.synchronize()
+        for got, exp in zip(a, expected):
+            assert torch.allclose(got.float(), exp.float(), rtol=tol,
+                                  atol=tol), \
+                (dtype, (got.float() - exp.float()).abs().max())
