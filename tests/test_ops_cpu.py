@@ -256,3 +256,57 @@ def test_noncontiguous_np2():
         except ValueError:
             pass
     """)
+
+
+def test_subset_ops_nonmember_root_np3():
+    """All collectives on a process set excluding global rank 0: the
+    coordinator participates only as a relay (lock-step frames)."""
+    run_workers(3, """
+        ps = hvd.add_process_set(hvd.ProcessSet([1, 2]))
+        if rank in (1, 2):
+            li = [1, 2].index(rank)
+            out = hvd.allreduce(torch.ones(4) * (rank), average=False,
+                                process_set=ps, name="s_ar")
+            assert out[0].item() == 3.0, out
+            g = hvd.allgather(torch.full((2, 2), float(rank)),
+                              process_set=ps, name="s_ag")
+            assert g.shape == (4, 2) and g[0, 0] == 1.0 and g[2, 0] == 2.0, g
+            b = hvd.broadcast(torch.ones(3) * rank, root_rank=2,
+                              process_set=ps, name="s_bc")
+            assert b[0].item() == 2.0, b
+            o, rs = hvd.alltoall(torch.arange(4).float() + rank * 10,
+                                 splits=[2, 2], process_set=ps, name="s_a2a")
+            exp = ([10.0, 11.0, 20.0, 21.0] if rank == 1
+                   else [12.0, 13.0, 22.0, 23.0])
+            assert o.tolist() == exp, (o, exp)
+            r = hvd.reducescatter(torch.arange(4).float(), op=hvd.Sum,
+                                  process_set=ps, name="s_rs")
+            exp = [0.0, 2.0] if rank == 1 else [4.0, 6.0]
+            assert r.tolist() == exp, r
+            hvd.barrier(process_set=ps)
+        # everyone joins a final global barrier
+        hvd.barrier()
+    """)
+
+
+def test_fusion_threshold_splitting_np2():
+    """Many tensors with a tiny fusion threshold: multiple fused responses
+    per cycle, all results exact."""
+    run_workers(2, """
+        ts = [torch.full((1000,), float(i + rank)) for i in range(40)]
+        outs = hvd.grouped_allreduce(ts, average=False, name="ft")
+        for i, o in enumerate(outs):
+            assert torch.allclose(o, torch.full((1000,), 2.0 * i + 1.0)), i
+    """, extra_env={"HOROVOD_FUSION_THRESHOLD": str(8 * 1024)})
+
+
+def test_allgather_broadcast_dtypes_np2():
+    run_workers(2, """
+        for dtype in [torch.uint8, torch.int64, torch.float16, torch.bool,
+                      torch.bfloat16]:
+            t = (torch.arange(4) % 2).to(dtype)
+            g = hvd.allgather(t, name=f"agd{dtype}")
+            assert g.shape[0] == 8, (dtype, g.shape)
+            b = hvd.broadcast(t, root_rank=0, name=f"bcd{dtype}")
+            assert torch.equal(b, t), dtype
+    """)
