@@ -12,6 +12,8 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_bf16.h>
 
+#include <algorithm>
+
 #include "kernels.h"
 
 namespace hvd {
@@ -53,8 +55,29 @@ __global__ __launch_bounds__(256) void adasum_dots_k(AdasumBatchArgs args,
   const T* __restrict__ b = (const T*)args.b[t];
   const long long n = (long long)args.numel[t];
   double d = 0, na = 0, nb = 0;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long long)gridDim.x * blockDim.x) {
+  // vectorized main loop: 16 B per lane per load (HBM-bound; guide G13)
+  constexpr int VS = 16 / (int)sizeof(T);
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  long long done = 0;
+  if ((((uintptr_t)a & 15) == 0) && (((uintptr_t)b & 15) == 0)) {
+    const long long nvec = n / VS;
+    done = nvec * VS;
+    for (long long i = tid; i < nvec; i += nthreads) {
+      union { uint4 u; T e[VS]; } va, vb;
+      va.u = ((const uint4*)a)[i];
+      vb.u = ((const uint4*)b)[i];
+#pragma unroll
+      for (int k = 0; k < VS; ++k) {
+        double av = to_d<T>(va.e[k]);
+        double bv = to_d<T>(vb.e[k]);
+        d += av * bv;
+        na += av * av;
+        nb += bv * bv;
+      }
+    }
+  }
+  for (long long i = done + tid; i < n; i += nthreads) {
     double av = to_d<T>(a[i]);
     double bv = to_d<T>(b[i]);
     d += av * bv;
@@ -97,8 +120,24 @@ __global__ __launch_bounds__(256) void adasum_scaledadd_k(AdasumBatchArgs args,
   double dot = dots[t * 3 + 0], na = dots[t * 3 + 1], nb = dots[t * 3 + 2];
   double ac = na > 0 ? 1.0 - dot / (2.0 * na) : 1.0;
   double bc = nb > 0 ? 1.0 - dot / (2.0 * nb) : 1.0;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long long)gridDim.x * blockDim.x) {
+  constexpr int VS = 16 / (int)sizeof(T);
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  long long done = 0;
+  if ((((uintptr_t)a & 15) == 0) && (((uintptr_t)b & 15) == 0)) {
+    const long long nvec = n / VS;
+    done = nvec * VS;
+    for (long long i = tid; i < nvec; i += nthreads) {
+      union { uint4 u; T e[VS]; } va, vb;
+      va.u = ((uint4*)a)[i];
+      vb.u = ((const uint4*)b)[i];
+#pragma unroll
+      for (int k = 0; k < VS; ++k)
+        va.e[k] = from_d<T>(ac * to_d<T>(va.e[k]) + bc * to_d<T>(vb.e[k]));
+      ((uint4*)a)[i] = va.u;
+    }
+  }
+  for (long long i = done + tid; i < n; i += nthreads) {
     a[i] = from_d<T>(ac * to_d<T>(a[i]) + bc * to_d<T>(b[i]));
   }
 }
@@ -106,8 +145,15 @@ __global__ __launch_bounds__(256) void adasum_scaledadd_k(AdasumBatchArgs args,
 template <typename T>
 hipError_t launch_adasum(const AdasumBatchArgs& args, double* dots, bool stage2,
                          hipStream_t stream) {
-  // stage2 reads dots; stage1 accumulates into them
-  dim3 grid(64, args.count), block(256);
+  // stage2 reads dots; stage1 accumulates into them.
+  // grid: fill the chip (256 CUs x 8 XCDs want >>256 workgroups; guide G1) —
+  // size x-dim to the largest tensor, stride the rest.
+  unsigned long long max_n = 0;
+  for (int i = 0; i < args.count; ++i)
+    if (args.numel[i] > max_n) max_n = args.numel[i];
+  int bx = (int)std::min<unsigned long long>(
+      2048, (max_n / (16 / sizeof(T)) + 255) / 256 + 1);
+  dim3 grid(bx, args.count), block(256);
   if (stage2)
     adasum_scaledadd_k<T><<<grid, block, 0, stream>>>(args, dots);
   else

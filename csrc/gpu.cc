@@ -8,6 +8,7 @@
 #include <c10/hip/HIPGuard.h>
 #include <c10/hip/HIPStream.h>
 
+#include <algorithm>
 #include <condition_variable>
 #include <cstring>
 #include <deque>
@@ -257,7 +258,10 @@ int64_t PackEntries(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
   int src_dt = 0, dst_dt = 0;
   auto flush = [&](int s_dt, int d_dt) {
     if (args.count == 0) return;
-    HIP_CHECK(BatchedCopyLaunch(args, s_dt, d_dt, any_scale, 32, ctx.stream.stream()));
+    // fill the chip: ~2048 workgroups total regardless of batch size
+    int bpc = std::min(256, std::max(8, 2048 / args.count));
+    HIP_CHECK(BatchedCopyLaunch(args, s_dt, d_dt, any_scale, bpc,
+                                ctx.stream.stream()));
     args.count = 0;
     any_scale = false;
   };
@@ -602,7 +606,7 @@ void Execute(GlobalState& st, Response& resp,
         args.dst[0] = e.output.data_ptr();
         args.numel[0] = (unsigned long long)e.output.numel();
         args.scale[0] = e.postscale;
-        HIP_CHECK(BatchedCopyLaunch(args, (int)wire, (int)wire, true, 32, stream));
+        HIP_CHECK(BatchedCopyLaunch(args, (int)wire, (int)wire, true, 256, stream));
       }
       break;
     }
