@@ -117,7 +117,20 @@ void FinalizerLoop() {
       op = std::move(g_finalizer.queue.front());
       g_finalizer.queue.pop_front();
     }
-    hipError_t e = hipEventSynchronize(op.done_event);
+    // low-latency completion: poll-yield first (hipEventSynchronize's
+    // blocking wait costs hundreds of us in wake latency), sleep once the
+    // op is clearly long
+    hipError_t e;
+    auto spin_start = std::chrono::steady_clock::now();
+    while ((e = hipEventQuery(op.done_event)) == hipErrorNotReady) {
+      if (std::chrono::duration<double>(std::chrono::steady_clock::now() -
+                                        spin_start)
+              .count() > 0.005) {
+        e = hipEventSynchronize(op.done_event);
+        break;
+      }
+      std::this_thread::yield();
+    }
     Status s = e == hipSuccess
                    ? Status::OK()
                    : Status::UnknownError(std::string("hipEventSynchronize: ") +
