@@ -103,6 +103,34 @@ std::mutex g_ctx_mu;
 std::unordered_map<int, DeviceCtx*> g_ctx;
 std::atomic<bool> g_rccl_used{false};
 
+// hipEvent pool (reference: GPUContext event pool with prepopulation,
+// hip_operations.cc:5-80) — create/destroy cost ~2 us each adds up at 161
+// events/step.
+std::mutex g_event_mu;
+std::vector<hipEvent_t> g_event_pool;
+
+hipEvent_t AcquireEvent() {
+  {
+    std::lock_guard<std::mutex> g(g_event_mu);
+    if (!g_event_pool.empty()) {
+      hipEvent_t ev = g_event_pool.back();
+      g_event_pool.pop_back();
+      return ev;
+    }
+  }
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  return ev;
+}
+
+void ReleaseEvent(hipEvent_t ev) {
+  std::lock_guard<std::mutex> g(g_event_mu);
+  if (g_event_pool.size() < 1024)
+    g_event_pool.push_back(ev);
+  else
+    hipEventDestroy(ev);
+}
+
 void FinalizerLoop() {
   while (true) {
     PendingOp op;
@@ -137,8 +165,8 @@ void FinalizerLoop() {
                                           hipGetErrorString(e));
     for (auto& entry : op.entries)
       if (entry.callback) entry.callback(s, entry);
-    hipEventDestroy(op.done_event);
-    for (auto ev : op.ready_events) hipEventDestroy(ev);
+    ReleaseEvent(op.done_event);
+    for (auto ev : op.ready_events) ReleaseEvent(ev);
     auto& st = State();
     auto tl = GetTimeline(st);
     if (tl && !op.entries.empty())
@@ -237,7 +265,7 @@ void Finalize(DeviceCtx& ctx, std::vector<TensorTableEntry> entries,
               std::vector<hipEvent_t> ready, const char* activity,
               int64_t start_us) {
   PendingOp op;
-  HIP_CHECK(hipEventCreateWithFlags(&op.done_event, hipEventDisableTiming));
+  op.done_event = AcquireEvent();
   HIP_CHECK(hipEventRecord(op.done_event, ctx.stream.stream()));
   op.entries = std::move(entries);
   op.ready_events = std::move(ready);
@@ -397,8 +425,7 @@ void ExecuteAdasum(GlobalState& st, DeviceCtx& ctx, Response& resp,
 }  // namespace
 
 uintptr_t RecordReadyEvent(int device) {
-  hipEvent_t ev;
-  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  hipEvent_t ev = AcquireEvent();
   HIP_CHECK(hipEventRecord(ev, c10::hip::getCurrentHIPStream(device).stream()));
   return (uintptr_t)ev;
 }
@@ -710,13 +737,18 @@ void Shutdown() {
   if (g_finalizer.started && g_finalizer.thread.joinable()) g_finalizer.thread.join();
   g_finalizer.started = false;
   g_finalizer.stop = false;
-  std::lock_guard<std::mutex> g(g_ctx_mu);
-  for (auto& kv : g_ctx) {
-    for (auto& ck : kv.second->comms) ncclCommDestroy(ck.second);
-    delete kv.second;
+  {
+    std::lock_guard<std::mutex> g(g_ctx_mu);
+    for (auto& kv : g_ctx) {
+      for (auto& ck : kv.second->comms) ncclCommDestroy(ck.second);
+      delete kv.second;
+    }
+    g_ctx.clear();
   }
-  g_ctx.clear();
   g_bootstrapped.clear();
+  std::lock_guard<std::mutex> g(g_event_mu);
+  for (auto ev : g_event_pool) hipEventDestroy(ev);
+  g_event_pool.clear();
 }
 
 }  // namespace gpu
