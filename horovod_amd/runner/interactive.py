@@ -4,7 +4,6 @@
 Reference: horovod.run / runner/__init__.py (interactive API used by
 test_interactiverun.py and the Spark layer).
 """
-import base64
 import os
 import subprocess
 import sys

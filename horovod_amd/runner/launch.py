@@ -8,11 +8,9 @@ multi-host uses ssh fan-out with the same env protocol.  No MPI.
 import argparse
 import os
 import shlex
-import signal
 import socket
 import subprocess
 import sys
-import time
 
 
 def find_free_port():

@@ -38,10 +38,8 @@ class ElasticSampler(torch.utils.data.Sampler):
         self.reset()
 
     def record_batch(self, batch_idx, batch_size):
-        """Record the indices of `batch_idx` as processed."""
-        start = self.rank * self.num_samples + batch_idx * batch_size
-        end = min(start + batch_size, (self.rank + 1) * self.num_samples)
-        # indices within this rank's shard
+        """Record the dataset indices of this rank's `batch_idx` as
+        processed (they are skipped after an elastic reset)."""
         shard = self.indices[batch_idx * batch_size:
                              (batch_idx + 1) * batch_size]
         self.processed_indices.update(shard)

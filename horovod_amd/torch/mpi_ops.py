@@ -7,7 +7,6 @@ same user surface on the MI355X-native core: one templated native entry point
 per op (no per-dtype symbols), wire-dtype compression fused into the pack
 kernels, RCCL over xGMI underneath.
 """
-import io
 from contextlib import contextmanager
 
 import torch

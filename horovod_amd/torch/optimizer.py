@@ -19,8 +19,6 @@ import torch
 from horovod_amd.torch.compression import Compression
 from horovod_amd.torch import mpi_ops
 from horovod_amd.torch.mpi_ops import (Adasum, Average, Sum,
-                                       allreduce_async_,
-                                       grouped_allreduce_async_,
                                        sparse_allreduce_async, size,
                                        synchronize)
 from horovod_amd.common.process_sets import global_process_set

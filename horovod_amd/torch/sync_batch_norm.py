@@ -11,8 +11,7 @@ import itertools
 import torch
 from torch.nn.modules.batchnorm import _BatchNorm
 
-from horovod_amd.torch.mpi_ops import (Sum, allreduce, size,
-                                       grouped_allreduce)
+from horovod_amd.torch.mpi_ops import allreduce, size
 from horovod_amd.common.process_sets import global_process_set
 
 

@@ -161,8 +161,7 @@ def test_reducescatter_np2():
 
 def test_reducescatter_uneven_np2():
     run_workers(2, """
-        t = torch.arange(6).float() * (rank + 1)  # first dim 6... wait 1-D len 6? use 5
-        t = torch.arange(5).float() * (rank + 1)
+        t = torch.arange(5).float() * (rank + 1)  # first dim 5: uneven over 2
         out = hvd.reducescatter(t, op=hvd.Sum, name="rsv")
         full = torch.arange(5).float() * 3
         if rank == 0:
