@@ -118,6 +118,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   // ---- completion ---------------------------------------------------------
   m.def("poll", [](int handle) { return State().handles.Poll(handle); });
+  m.def("flush", [] { FlushCycle(); });
   m.def("wait", &WaitHandle);
 
   m.def("adasum_combine_", [](std::vector<at::Tensor> a,

@@ -33,7 +33,6 @@ Status TensorQueue::Add(Request req, TensorTableEntry entry) {
   }
   table_.emplace(std::move(key), std::move(entry));
   messages_.push_back(std::move(req));
-  cv_.notify_one();
   return Status::OK();
 }
 
@@ -49,19 +48,12 @@ Status TensorQueue::AddMulti(std::vector<Request>& reqs,
                    std::move(entries[i]));
     messages_.push_back(std::move(reqs[i]));
   }
-  cv_.notify_one();
   return Status::OK();
 }
 
 bool TensorQueue::has_messages() const {
   std::lock_guard<std::mutex> g(mu_);
   return !messages_.empty();
-}
-
-void TensorQueue::WaitForMessages(double ms) {
-  std::unique_lock<std::mutex> lk(mu_);
-  cv_.wait_for(lk, std::chrono::duration<double, std::milli>(ms),
-               [&] { return !messages_.empty(); });
 }
 
 std::vector<Request> TensorQueue::PopMessages() {
@@ -177,6 +169,15 @@ void SetTimeline(GlobalState& st, std::shared_ptr<Timeline> t) {
 }
 
 bool IsInitialized() { return State().initialized; }
+
+void FlushCycle() {
+  auto& st = State();
+  {
+    std::lock_guard<std::mutex> g(st.pace_mu);
+    st.flush_requested = true;
+  }
+  st.pace_cv.notify_one();
+}
 
 namespace {
 
@@ -641,13 +642,21 @@ void BackgroundLoop(GlobalState& st) {
     }
     for (auto& resp : rl.responses) PerformOperation(st, resp);
     if (rl.shutdown) break;
-    if (rl.responses.empty() && !st.queue.has_messages() &&
-        !st.controller->has_pending()) {
+    // pace the cycle cadence: gradient bursts accumulate for up to
+    // cycle_time so they fuse into large buckets; a flush (synchronize())
+    // cuts the wait for the step's trailing bucket.
+    {
       auto elapsed = std::chrono::duration<double, std::milli>(
                          std::chrono::steady_clock::now() - cycle_start)
                          .count();
       double remain = st.controller->cycle_time_ms() - elapsed;
-      if (remain > 0) st.queue.WaitForMessages(remain);
+      if (remain > 0) {
+        std::unique_lock<std::mutex> lk(st.pace_mu);
+        st.pace_cv.wait_for(lk,
+                            std::chrono::duration<double, std::milli>(remain),
+                            [&] { return st.flush_requested; });
+        st.flush_requested = false;
+      }
     }
   }
   gpu::WaitAllPending();

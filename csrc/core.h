@@ -37,9 +37,6 @@ class TensorQueue {
   void FailAll(const Status& s);
   size_t size() const;
   bool has_messages() const;
-  // Sleep up to `ms`, waking immediately when a new request is enqueued —
-  // the trailing gradient bucket of a step never waits out a cycle.
-  void WaitForMessages(double ms);
 
  private:
   mutable std::mutex mu_;
@@ -98,6 +95,13 @@ struct GlobalState {
   HandleManager handles;
   std::thread bg_thread;
 
+  // cycle pacing: cycles are paced at cycle_time so gradient bursts fuse
+  // into large buckets; flush() (called from synchronize()) cuts the wait
+  // so the trailing bucket of a step fires immediately.
+  std::mutex pace_mu;
+  std::condition_variable pace_cv;
+  bool flush_requested = false;
+
   // join(): this rank's joined state + zero-substitute device, per set.
   std::mutex join_mu;
   std::unordered_map<int32_t, bool> local_joined;
@@ -119,6 +123,9 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
                  int port, const ControllerConfig& cfg);
 void ShutdownHorovod();
 bool IsInitialized();
+// Cut the current cycle-pacing wait (the caller is about to block on
+// results): lets the background thread negotiate the tail immediately.
+void FlushCycle();
 
 // Enqueue API (reference: EnqueueTensor* operations.cc:1408-2057) -----------
 // wire_dtype may differ from the tensors' dtype: the fusion pack kernel

@@ -160,6 +160,7 @@ def poll(handle):
 def synchronize(handle):
     """Wait for the async op and return its output tensor(s)."""
     info = _handles.pop(handle, None)
+    _core.flush()  # cut the cycle-pacing window: we are about to block
     outs, extra, result_int = _translate_error(_core.wait, handle)
     if info is not None and info.kind == "join":
         return result_int
