@@ -41,6 +41,9 @@ def parse_args():
                    help="disable backward/allreduce overlap (ablation)")
     p.add_argument("--fused-sgd", action="store_true",
                    help="use the CDNA4 fused SGD step kernel")
+    p.add_argument("--persistent-grads", action="store_true",
+                   help="zero_grad(set_to_none=False): keep gradient buffers "
+                        "allocated across steps")
     p.add_argument("--hipgraph", action="store_true",
                    help="capture forward+backward in a hipGraph; allreduce + "
                         "optimizer run after replay (trades overlap for "
