@@ -165,12 +165,6 @@ class _DistributedOptimizer(torch.optim.Optimizer):
     def synchronize(self):
         """Wait for all outstanding gradient allreduces (reference:
         optimizer.py:255-323)."""
-        completed = set()
-        # fire any params whose hook never ran but are due (missing grads)
-        for p in self._requires_update:
-            if p not in self._handles and self._allreduce_delay.get(p, 1) == \
-                    self.backward_passes_per_step:
-                continue  # no backward happened at all for p this round
         for p, handle in list(self._handles.items()):
             if handle is None:
                 # group member whose group never fired: fire it now alone
@@ -189,7 +183,6 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             else:
                 synchronize(handle)
             self._allreduce_delay[p] = self.backward_passes_per_step
-            completed.add(p)
         self._handles.clear()
         self._synchronized = True
 
