@@ -114,11 +114,6 @@ void StarComm::Shutdown() {
   alive_ = false;
 }
 
-int StarComm::FdFor(int peer_rank) const {
-  if (rank_ == 0) return fds_[peer_rank];
-  return fds_[0];
-}
-
 void StarComm::SendRaw(int fd, const void* data, size_t len) {
   const char* p = (const char*)data;
   while (len > 0) {
@@ -217,19 +212,6 @@ std::string StarComm::ScatterFrames(const std::vector<std::string>& frames) {
     return frames.empty() ? std::string() : frames[0];
   }
   return RecvFrame(fds_[0]);
-}
-
-void StarComm::SendToRank(int dst, const void* data, size_t len) {
-  SendRaw(fds_[dst], data, len);
-}
-void StarComm::RecvFromRank(int src, void* data, size_t len) {
-  RecvRaw(fds_[src], data, len);
-}
-void StarComm::SendToRoot(const void* data, size_t len) {
-  SendRaw(fds_[0], data, len);
-}
-void StarComm::RecvFromRoot(void* data, size_t len) {
-  RecvRaw(fds_[0], data, len);
 }
 
 }  // namespace hvd

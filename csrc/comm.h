@@ -44,14 +44,6 @@ class StarComm {
   // Root sends frames[r] to each rank r; every rank returns its own frame.
   std::string ScatterFrames(const std::vector<std::string>& frames);
 
-  // Point-to-point, used by the CPU data plane (star-routed through root).
-  // Root side:
-  void SendToRank(int dst, const void* data, size_t len);
-  void RecvFromRank(int src, void* data, size_t len);
-  // Worker side:
-  void SendToRoot(const void* data, size_t len);
-  void RecvFromRoot(void* data, size_t len);
-
   bool alive() const { return alive_; }
 
  private:
@@ -59,7 +51,6 @@ class StarComm {
   std::string RecvFrame(int fd);
   void SendRaw(int fd, const void* data, size_t len);
   void RecvRaw(int fd, void* data, size_t len);
-  int FdFor(int peer_rank) const;
 
   int rank_ = 0;
   int size_ = 1;

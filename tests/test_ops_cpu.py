@@ -309,3 +309,28 @@ def test_allgather_broadcast_dtypes_np2():
             b = hvd.broadcast(t, root_rank=0, name=f"bcd{dtype}")
             assert torch.equal(b, t), dtype
     """)
+
+
+def test_remove_process_set_np2():
+    run_workers(2, """
+        ps = hvd.add_process_set(hvd.ProcessSet([0, 1]))
+        out = hvd.allreduce(torch.ones(2), average=False, process_set=ps,
+                            name="rps1")
+        assert out.sum().item() == 4.0
+        hvd.remove_process_set(ps)
+        ps2 = hvd.add_process_set(hvd.ProcessSet([0, 1]))
+        out = hvd.allreduce(torch.ones(2), average=False, process_set=ps2,
+                            name="rps2")
+        assert out.sum().item() == 4.0
+    """)
+
+
+def test_cache_invalidation_reshape_np2():
+    """Same tensor name with a changed shape: the cache entry must be
+    invalidated and renegotiated, not reused."""
+    run_workers(2, """
+        for shape in [(8,), (8,), (16,), (16,), (8,)]:
+            out = hvd.allreduce(torch.ones(*shape), average=False,
+                                name="reshape_me")
+            assert out.sum().item() == 2.0 * shape[0], (shape, out.sum())
+    """)
