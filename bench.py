@@ -37,8 +37,6 @@ def parse_args():
     p.add_argument("--use-adasum", action="store_true")
     p.add_argument("--no-bf16", action="store_true",
                    help="disable bf16 autocast (fp32 compute)")
-    p.add_argument("--no-overlap", action="store_true",
-                   help="disable backward/allreduce overlap (ablation)")
     p.add_argument("--fused-sgd", action="store_true",
                    help="use the CDNA4 fused SGD step kernel")
     p.add_argument("--persistent-grads", action="store_true",

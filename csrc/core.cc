@@ -615,6 +615,8 @@ void SetThreadAffinity() {
 
 void BackgroundLoop(GlobalState& st) {
   SetThreadAffinity();
+  const bool mark_cycles = std::getenv("HOROVOD_TIMELINE_MARK_CYCLES") != nullptr;
+  uint64_t cycle_no = 0;
   // Shutdown protocol: a graceful exit needs every rank's agreement (AND bit
   // in the cache-sync round) so no peer is left mid-collective; but a rank
   // must never hang forever on a dead peer — after the grace window it
@@ -641,6 +643,11 @@ void BackgroundLoop(GlobalState& st) {
       break;
     }
     for (auto& resp : rl.responses) PerformOperation(st, resp);
+    ++cycle_no;
+    if (mark_cycles && !rl.responses.empty()) {
+      auto tl = GetTimeline(st);
+      if (tl) tl->Marker("CYCLE_" + std::to_string(cycle_no));
+    }
     if (rl.shutdown) break;
     // pace the cycle cadence: gradient bursts accumulate for up to
     // cycle_time so they fuse into large buckets; a flush (synchronize())

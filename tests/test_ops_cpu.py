@@ -334,3 +334,63 @@ def test_cache_invalidation_reshape_np2():
                                 name="reshape_me")
             assert out.sum().item() == 2.0 * shape[0], (shape, out.sum())
     """)
+
+
+def test_grouped_allgather_reducescatter_np2():
+    run_workers(2, """
+        ts = [torch.full((2, 3), float(rank)), torch.full((1, 3), float(rank))]
+        outs = hvd.grouped_allgather(ts, name="gag")
+        assert outs[0].shape == (4, 3) and outs[1].shape == (2, 3)
+        assert outs[0][0, 0] == 0.0 and outs[0][2, 0] == 1.0
+        rs_in = [torch.ones(4, 2) * (rank + 1), torch.ones(2, 2) * (rank + 1)]
+        routs = hvd.grouped_reducescatter(rs_in, op=hvd.Sum, name="grs")
+        assert routs[0].shape == (2, 2) and routs[0][0, 0] == 3.0
+        assert routs[1].shape == (1, 2) and routs[1][0, 0] == 3.0
+    """)
+
+
+def test_reducescatter_default_average_np2():
+    run_workers(2, """
+        t = torch.full((4,), 2.0 * (rank + 1))
+        out = hvd.reducescatter(t, name="rsavg")  # default = Average
+        assert torch.allclose(out, torch.full((2,), 3.0)), out
+    """)
+
+
+def test_scalar_and_edge_shapes_np2():
+    run_workers(2, """
+        # 0-dim tensor allreduce
+        s = torch.tensor(float(rank + 1))
+        out = hvd.allreduce(s, average=False, name="scalar")
+        assert out.item() == 3.0
+        # single-element broadcast
+        b = torch.tensor([float(rank)])
+        hvd.broadcast_(b, root_rank=1, name="sb")
+        assert b.item() == 1.0
+        # empty-ish allgather (one rank contributes zero rows)
+        t = torch.ones(0 if rank == 0 else 2, 3)
+        g = hvd.allgather(t, name="eag")
+        assert g.shape == (2, 3), g.shape
+    """)
+
+
+def test_join_last_rank_np3():
+    run_workers(3, """
+        import time
+        time.sleep(0.2 * rank)  # rank 2 joins last
+        last = hvd.join()
+        assert last == 2, last
+    """)
+
+
+def test_timeline_mark_cycles_np2(tmp_path):
+    import json as _json
+    tl = str(tmp_path / "tlm.json")
+    run_workers(2, f"""
+        hvd.start_timeline({tl!r})
+        for i in range(5):
+            hvd.allreduce(torch.ones(10), average=False, name=f"m{{i}}")
+        hvd.stop_timeline()
+    """, extra_env={"HOROVOD_TIMELINE_MARK_CYCLES": "1"})
+    events = _json.load(open(tl))
+    assert any(e.get("name", "").startswith("CYCLE_") for e in events)
