@@ -221,6 +221,17 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     }
   }
 
+  // re-classify pending hits whose cache slot was since evicted/replaced
+  // (capacity-pressure Put evictions would otherwise strand them forever)
+  for (auto it = cached_pending_.begin(); it != cached_pending_.end();) {
+    if (cache_.Lookup(*it) != ResponseCache::State::HIT) {
+      slow.push_back(*it);
+      it = cached_pending_.erase(it);
+    } else {
+      ++it;
+    }
+  }
+
   // -- 2. bitvector rounds --------------------------------------------------
   const size_t nbits = cfg_.cache_capacity;
   const size_t nbytes = (nbits + 7) / 8;
@@ -449,16 +460,31 @@ std::vector<Response> Controller::CoordinatorProcess(
     }
     if (pt.count >= needed) {
       Response r = ConstructResponse(it->first, pt);
-      if (!r.group_key.empty() && r.group_size > 1 &&
-          r.type != ResponseType::ERROR) {
-        // strict group semantics (reference GroupTable): hold until every
-        // member of the group is ready, then emit ONE fused response.
-        auto& hold = group_hold_[SetKey(r.process_set_id, r.group_key)];
-        hold.push_back(std::move(r));
-        if ((int32_t)hold.size() == hold.front().group_size) {
-          ready_responses_.push_back(MergeGroup(hold));
-          group_hold_.erase(SetKey(hold.front().process_set_id,
-                                   hold.front().group_key));
+      if (!r.group_key.empty() && r.group_size > 1) {
+        std::string gk = SetKey(r.process_set_id, r.group_key);
+        if (r.type == ResponseType::ERROR) {
+          // poisoned group: release anything already held as individual
+          // responses so peers error out instead of hanging on the merge
+          poisoned_groups_.insert(gk);
+          auto hit = group_hold_.find(gk);
+          if (hit != group_hold_.end()) {
+            for (auto& held : hit->second)
+              ready_responses_.push_back(std::move(held));
+            group_hold_.erase(hit);
+          }
+          ready_responses_.push_back(std::move(r));
+        } else if (poisoned_groups_.count(gk)) {
+          ready_responses_.push_back(std::move(r));
+        } else {
+          // strict group semantics (reference GroupTable): hold until every
+          // member of the group is ready, then emit ONE fused response.
+          auto& hold = group_hold_[gk];
+          hold.push_back(std::move(r));
+          if ((int32_t)hold.size() == hold.front().group_size) {
+            ready_responses_.push_back(MergeGroup(hold));
+            group_hold_.erase(gk);
+            poisoned_groups_.erase(gk);
+          }
         }
       } else {
         ready_responses_.push_back(std::move(r));

@@ -167,6 +167,7 @@ class Controller {
   std::deque<Response> ready_responses_;  // completed, awaiting fusion window
   // grouped-op holding area: set:group_key -> completed singles
   std::unordered_map<std::string, std::vector<Response>> group_hold_;
+  std::unordered_set<std::string> poisoned_groups_;
 
   std::unordered_map<int32_t, ProcessSetInfo> process_sets_;
   int32_t next_set_id_ = 1;

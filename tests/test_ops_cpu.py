@@ -394,3 +394,16 @@ def test_timeline_mark_cycles_np2(tmp_path):
     """, extra_env={"HOROVOD_TIMELINE_MARK_CYCLES": "1"})
     events = _json.load(open(tl))
     assert any(e.get("name", "").startswith("CYCLE_") for e in events)
+
+
+def test_grouped_error_does_not_hang_np2():
+    """One member of a group has mismatched shapes: all ranks must get an
+    error (or complete), never hang."""
+    run_workers(2, """
+        ts = [torch.ones(4), torch.ones(6 if rank == 0 else 7)]
+        try:
+            hvd.grouped_allreduce(ts, average=False, name="poison")
+            raise SystemExit("expected an error")
+        except RuntimeError as e:
+            assert "Mismatched" in str(e), e
+    """, timeout=90)
