@@ -46,7 +46,6 @@ class Autotuner {
   Params best_;
   double best_score_ = -1.0;
   bool done_ = false;
-  bool warmup_done_ = false;
 
   std::vector<std::array<double, 2>> xs_;  // normalized samples
   std::vector<double> ys_;                 // scores (normalized later)

@@ -137,7 +137,7 @@ void ReleaseEvent(hipEvent_t ev) {
   if (g_event_pool.size() < 1024)
     g_event_pool.push_back(ev);
   else
-    hipEventDestroy(ev);
+    (void)hipEventDestroy(ev);
 }
 
 void FinalizerLoop() {
@@ -771,7 +771,7 @@ void Shutdown() {
   }
   g_bootstrapped.clear();
   std::lock_guard<std::mutex> g(g_event_mu);
-  for (auto ev : g_event_pool) hipEventDestroy(ev);
+  for (auto ev : g_event_pool) (void)hipEventDestroy(ev);
   g_event_pool.clear();
 }
 

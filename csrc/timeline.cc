@@ -30,7 +30,7 @@ std::string Escape(const std::string& s) {
 }  // namespace
 
 Timeline::Timeline(const std::string& path, int rank)
-    : t0_(std::chrono::steady_clock::now()), rank_(rank) {
+    : t0_(std::chrono::steady_clock::now()) {
   std::string p = path;
   if (rank != 0) p += "." + std::to_string(rank);
   file_ = std::fopen(p.c_str(), "w");

@@ -22,6 +22,7 @@ namespace hvd {
 
 class Timeline {
  public:
+  // rank names the output file (rank 0 writes `path`, others `path.<rank>`)
   Timeline(const std::string& path, int rank);
   ~Timeline();
 
@@ -50,7 +51,6 @@ class Timeline {
 
   std::chrono::steady_clock::time_point t0_;
   FILE* file_ = nullptr;
-  int rank_;
   std::mutex mu_;
   std::condition_variable cv_;
   std::deque<Record> queue_;
