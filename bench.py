@@ -37,8 +37,10 @@ def parse_args():
     p.add_argument("--use-adasum", action="store_true")
     p.add_argument("--no-bf16", action="store_true",
                    help="disable bf16 autocast (fp32 compute)")
-    p.add_argument("--fused-sgd", action="store_true",
-                   help="use the CDNA4 fused SGD step kernel")
+    p.add_argument("--fused-sgd", dest="fused_sgd", action="store_true",
+                   default=True,
+                   help="use the CDNA4 fused SGD step kernel (default)")
+    p.add_argument("--no-fused-sgd", dest="fused_sgd", action="store_false")
     p.add_argument("--persistent-grads", action="store_true",
                    help="zero_grad(set_to_none=False): keep gradient buffers "
                         "allocated across steps")

@@ -295,3 +295,67 @@ def test_adasum_optimizer_np2():
         assert torch.allclose(g[0], g[1], atol=1e-6), \
             (g[0] - g[1]).abs().max()
     """)
+
+
+def test_optimizer_with_process_set_np3():
+    """DistributedOptimizer over a 2-rank subset: members train in lockstep,
+    the outsider trains independently."""
+    run_workers(3, """
+        ps = hvd.add_process_set(hvd.ProcessSet([0, 1]))
+        torch.manual_seed(4)
+        model = torch.nn.Linear(5, 1)
+        if rank in (0, 1):
+            opt = hvd.DistributedOptimizer(
+                torch.optim.SGD(model.parameters(), lr=0.1),
+                named_parameters=model.named_parameters(), process_set=ps)
+            hvd.broadcast_parameters(model.state_dict(), root_rank=0,
+                                     process_set=ps)
+            torch.manual_seed(10 + rank)
+            for _ in range(3):
+                opt.zero_grad()
+                model(torch.randn(4, 5)).sum().backward()
+                opt.step()
+            flat = torch.cat([p.detach().flatten()
+                              for p in model.parameters()])
+            g = hvd.allgather(flat.unsqueeze(0), process_set=ps, name="sub_p")
+            assert torch.allclose(g[0], g[1], atol=1e-6)
+        hvd.barrier()
+    """)
+
+
+def test_mismatched_dtype_and_op_errors_np2():
+    run_workers(2, """
+        t = torch.ones(4) if rank == 0 else torch.ones(4, dtype=torch.float64)
+        try:
+            hvd.allreduce(t, average=False, name="dt_bad")
+            raise SystemExit("expected dtype mismatch error")
+        except RuntimeError as e:
+            assert "data types" in str(e) or "Mismatched" in str(e), e
+        op = hvd.Sum if rank == 0 else hvd.Max
+        try:
+            hvd.allreduce(torch.ones(4), op=op, name="op_bad")
+            raise SystemExit("expected op mismatch error")
+        except RuntimeError as e:
+            assert "Mismatched" in str(e), e
+    """)
+
+
+def test_gradient_predivide_np2():
+    run_workers(2, """
+        torch.manual_seed(2)
+        model = torch.nn.Linear(3, 1)
+        opt = hvd.DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1),
+            named_parameters=model.named_parameters(),
+            gradient_predivide_factor=2.0)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        ref = [p.detach().clone() for p in model.parameters()]
+        x = torch.ones(2, 3)
+        opt.zero_grad()
+        model(x).sum().backward()
+        # expected grad: averaged over 2 identical ranks == local grad
+        expected = [p.grad.clone() for p in model.parameters()]
+        opt.step()
+        for p, r, e in zip(model.parameters(), ref, expected):
+            assert torch.allclose(p.detach(), r - 0.1 * e, atol=1e-6)
+    """)
