@@ -407,3 +407,32 @@ def test_grouped_error_does_not_hang_np2():
         except RuntimeError as e:
             assert "Mismatched" in str(e), e
     """, timeout=90)
+
+
+def test_exhaustive_op_dtype_sweep_np2():
+    """Reference-style exhaustive sweep: every reduce op x dtype x shape
+    (test_torch.py's coverage model, compressed into one launch)."""
+    run_workers(2, """
+        import itertools
+        ops = [("sum", hvd.Sum), ("min", hvd.Min), ("max", hvd.Max),
+               ("prod", hvd.Product)]
+        dtypes = [torch.float32, torch.float64, torch.float16,
+                  torch.bfloat16, torch.int32, torch.int64, torch.uint8]
+        shapes = [(1,), (17, 3), (128,)]
+        for (opname, op), dtype, shape in itertools.product(ops, dtypes,
+                                                            shapes):
+            base = (torch.arange(int(torch.tensor(shape).prod()))
+                    .reshape(shape) % 5 + 1)
+            a = base.to(dtype)                  # rank 0 contribution
+            b = (base * 2).clamp(max=9).to(dtype)  # rank 1 contribution
+            mine = a if rank == 0 else b
+            out = hvd.allreduce(mine, op=op,
+                                name=f"sweep.{opname}.{dtype}.{len(shape)}")
+            af, bf = a.double(), b.double()
+            expected = {"sum": af + bf, "min": torch.minimum(af, bf),
+                        "max": torch.maximum(af, bf),
+                        "prod": af * bf}[opname]
+            assert torch.allclose(out.double(), expected, rtol=1e-2,
+                                  atol=1e-2), (opname, dtype, shape,
+                                               out, expected)
+    """, timeout=300)
