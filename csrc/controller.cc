@@ -317,11 +317,20 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
       return r;
     };
     for (int s : hit_slots) fast.push_back(consume(s));
+    // groups in deterministic order (min slot): unordered_map iteration
+    // order differs across ranks, and the response stream must be
+    // bit-identical everywhere (it serializes the RCCL call order)
+    std::vector<std::vector<int>*> ordered_groups;
     for (auto& kv : group_slots) {
-      auto& slots = kv.second;
+      std::sort(kv.second.begin(), kv.second.end());
+      ordered_groups.push_back(&kv.second);
+    }
+    std::sort(ordered_groups.begin(), ordered_groups.end(),
+              [](auto* a, auto* b) { return (*a)[0] < (*b)[0]; });
+    for (auto* slots_p : ordered_groups) {
+      auto& slots = *slots_p;
       if ((int32_t)slots.size() != cache_.Get(slots[0]).group_size)
         continue;  // incomplete group: stays pending, re-votes next cycle
-      std::sort(slots.begin(), slots.end());
       std::vector<Response> singles;
       for (int s : slots) singles.push_back(consume(s));
       fast.push_back(MergeGroup(singles));
