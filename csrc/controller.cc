@@ -106,7 +106,8 @@ void ResponseCache::Evict(int slot) {
 
 void ResponseCache::Put(const Response& response, const std::vector<Request>& reqs) {
   if (response.type == ResponseType::JOIN || response.type == ResponseType::BARRIER ||
-      response.type == ResponseType::ERROR || response.type == ResponseType::TUNE)
+      response.type == ResponseType::ERROR || response.type == ResponseType::TUNE ||
+      response.type == ResponseType::ALLTOALL)  // alltoall renegotiates
     return;
   // Split a fused response into single-tensor cache entries.
   size_t nsizes_per = response.names.size()
@@ -378,7 +379,9 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
       auto key = SetKey(req.process_set_id, req.name);
       PendingTensor pt;
       pt.requests = {req};
-      ready.push_back(ConstructResponse(key, pt));
+      Response r = ConstructResponse(key, pt);
+      if (r.type == ResponseType::JOIN) r.last_joined_rank = rank_;
+      ready.push_back(std::move(r));
     }
     for (auto& req : cached_pending_) {
       PendingTensor pt;
