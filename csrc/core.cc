@@ -523,9 +523,14 @@ void PerformOperation(GlobalState& st, Response& resp) {
     }
   } else if (member && joined &&
              (resp.type == ResponseType::ALLREDUCE ||
-              resp.type == ResponseType::ADASUM)) {
+              resp.type == ResponseType::ADASUM ||
+              resp.type == ResponseType::ALLGATHER ||
+              resp.type == ResponseType::BROADCAST)) {
     // Zero substitution for a joined rank (reference: tensor_queue.cc
-    // GetTensorEntriesFromResponse join path).
+    // GetTensorEntriesFromResponse join path): allreduce/adasum contribute
+    // zeros, allgather contributes zero rows, broadcast receives into a
+    // scratch buffer — all so the rank still participates in the RCCL
+    // collective its comm peers will issue.
     auto shapes = ParseShapes(resp);
     int dev;
     {
@@ -541,7 +546,10 @@ void PerformOperation(GlobalState& st, Response& resp) {
                       .dtype(DataTypeToTorch(resp.dtype))
                       .device(dev == CPU_DEVICE_ID ? at::Device(at::kCPU)
                                                    : at::Device(at::kCUDA, dev));
-      e.tensor = at::zeros(shapes[i], opts);
+      std::vector<int64_t> shape = shapes[i];
+      if (resp.type == ResponseType::ALLGATHER && !shape.empty())
+        shape[0] = 0;  // contribute zero rows
+      e.tensor = at::zeros(shape, opts);
       e.device = dev;
       e.process_set_id = resp.process_set_id;
       entries.push_back(std::move(e));

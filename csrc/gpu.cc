@@ -462,6 +462,12 @@ void Execute(GlobalState& st, Response& resp,
     g_bootstrapped[resp.process_set_id] = true;
   }
   if (entries.empty()) return;  // relay-only rank (not a member)
+  if (entries.size() != resp.names.size())
+    throw std::runtime_error(
+        "horovod_amd: fused response entry/name count mismatch (" +
+        std::to_string(entries.size()) + " vs " +
+        std::to_string(resp.names.size()) +
+        ") — fusion layout would diverge across ranks");
   int device = entries[0].device;
   auto& ctx = GetCtx(device);
   ncclComm_t comm = EnsureComm(st, ctx, resp.process_set_id);
