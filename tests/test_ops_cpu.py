@@ -436,3 +436,15 @@ def test_exhaustive_op_dtype_sweep_np2():
                                   atol=1e-2), (opname, dtype, shape,
                                                out, expected)
     """, timeout=300)
+
+
+def test_join_with_allgather_np2():
+    run_workers(2, """
+        if rank == 0:
+            g = hvd.allgather(torch.full((2, 3), 7.0), name="jag")
+            # rank 1 already joined: only our rows come back
+            assert g.shape == (2, 3), g.shape
+            out = hvd.allreduce(torch.ones(4), average=False, name="jar")
+            assert out.sum().item() == 4.0  # peer contributes zeros
+        hvd.join()
+    """, timeout=120)
