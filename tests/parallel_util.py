@@ -10,7 +10,7 @@ import textwrap
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def run_workers(np, body, timeout=180, extra_env=None):
+def run_workers(np, body, timeout=300, extra_env=None):
     """Run `body` (python source; has `hvd`, `torch`, `rank`, `size` in
     scope) in np processes.  Raises on nonzero exits; returns list of stdout
     strings by rank."""
