@@ -41,6 +41,8 @@ def parse_args():
                    default=True,
                    help="use the CDNA4 fused SGD step kernel (default)")
     p.add_argument("--no-fused-sgd", dest="fused_sgd", action="store_false")
+    p.add_argument("--fused-bn", action="store_true",
+                   help="use the fused CDNA4 BatchNorm(+Add)+ReLU kernels")
     p.add_argument("--persistent-grads", action="store_true",
                    help="zero_grad(set_to_none=False): keep gradient buffers "
                         "allocated across steps")
@@ -71,7 +73,8 @@ def main():
                 "resnet152": resnet152, "bert-large": bert_large,
                 "bert-base": bert_base}[args.model]
     torch.manual_seed(42)
-    model = model_fn().to(device)
+    model = (model_fn(fused_bn=True) if (args.fused_bn and not is_bert)
+             else model_fn()).to(device)
     if cuda and not is_bert:
         model = model.to(memory_format=torch.channels_last)
 
@@ -212,6 +215,7 @@ def main():
                 "compression": args.compression,
                 "reduction": "adasum" if args.use_adasum else "average",
                 "fused_sgd": args.fused_sgd,
+                "fused_bn": args.fused_bn,
                 "hipgraph": args.hipgraph,
             },
         }

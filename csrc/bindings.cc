@@ -135,6 +135,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                  weight_decay, dampening, nesterov);
         });
 
+  m.def("fused_bn_relu_forward",
+        [](at::Tensor x, py::object residual, at::Tensor gamma, at::Tensor beta,
+           at::Tensor rm, at::Tensor rv, double momentum, double eps) {
+          at::Tensor res;
+          if (!residual.is_none()) res = residual.cast<at::Tensor>();
+          return hvd::gpu::FusedBnReluForward(x, res, gamma, beta, rm, rv,
+                                              momentum, eps);
+        });
+  m.def("fused_bn_relu_backward",
+        [](at::Tensor x, at::Tensor y, at::Tensor dy, at::Tensor mean,
+           at::Tensor invstd, at::Tensor gamma, bool need_res) {
+          return hvd::gpu::FusedBnReluBackward(x, y, dy, mean, invstd, gamma,
+                                               need_res);
+        });
+
   // ---- timeline -----------------------------------------------------------
   m.def("start_timeline", [](const std::string& path, bool mark_cycles) {
     auto& st = State();

@@ -1,0 +1,302 @@
+// Fused BatchNorm(+Add)+ReLU kernels for CDNA4, NHWC (channels_last).
+//
+// Motivation (measured, profiles/pipeline_study.md): MIOpen's split
+// BN-forward / BN-backward plus the separate ReLU / residual-add elementwise
+// passes are ~30% of a ResNet-50 bf16 step's GPU time.  Fusing
+// normalize+affine+add+relu into one HBM pass per tensor (and the backward
+// reductions into one) removes whole read/write passes.
+//
+// Layout contract: activations are channels_last-dense [count=N*H*W, C]
+// with C contiguous and C % 8 == 0 (every ResNet channel width).  Stats
+// accumulate in fp32 via per-block LDS bins (C <= 4096 -> <= 32 KB LDS)
+// flushed with one global atomicAdd per channel per block (guide G12).
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
+
+#include <algorithm>
+
+#include "kernels.h"
+
+namespace hvd {
+namespace gpu {
+
+namespace {
+
+template <typename T>
+__device__ __forceinline__ float to_f(T v) { return (float)v; }
+template <>
+__device__ __forceinline__ float to_f<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+template <typename T>
+__device__ __forceinline__ T from_f(float v) { return (T)v; }
+template <>
+__device__ __forceinline__ __hip_bfloat16 from_f<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+
+// ---- forward stats: per-channel sum and sum-of-squares --------------------
+template <typename T>
+__global__ __launch_bounds__(256) void bn_stats_k(const T* __restrict__ x,
+                                                  long long total, int C,
+                                                  float* __restrict__ sums,
+                                                  float* __restrict__ sqs) {
+  extern __shared__ float lds[];  // [C] sum, [C] sq
+  float* lsum = lds;
+  float* lsq = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  // 8 contiguous elements per iteration (16 B for bf16/fp16)
+  const long long nvec = total / 8;
+  for (long long i = tid; i < nvec; i += nthreads) {
+    union { uint4 u; T e[8]; } v;
+    // T = float: 8 floats = 32 B -> two uint4 loads
+    if (sizeof(T) == 2) {
+      v.u = ((const uint4*)x)[i];
+    } else {
+      ((uint4*)&v)[0] = ((const uint4*)x)[i * 2];
+      ((uint4*)&v)[1] = ((const uint4*)x)[i * 2 + 1];
+    }
+    int c0 = (int)((i * 8) % C);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f = to_f<T>(v.e[k]);
+      atomicAdd(&lsum[c0 + k], f);
+      atomicAdd(&lsq[c0 + k], f * f);
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lsum[c] != 0.f || lsq[c] != 0.f) {
+      atomicAdd(&sums[c], lsum[c]);
+      atomicAdd(&sqs[c], lsq[c]);
+    }
+  }
+}
+
+// ---- forward apply: y = relu(gamma*xhat + beta [+ residual]) --------------
+template <typename T, bool ADD>
+__global__ __launch_bounds__(256) void bn_apply_relu_k(
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    long long total, int C) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const long long nvec = total / 8;
+  for (long long i = tid; i < nvec; i += nthreads) {
+    union { uint4 u; T e[8]; } vx, vr, vy;
+    if (sizeof(T) == 2) {
+      vx.u = ((const uint4*)x)[i];
+      if (ADD) vr.u = ((const uint4*)res)[i];
+    } else {
+      ((uint4*)&vx)[0] = ((const uint4*)x)[i * 2];
+      ((uint4*)&vx)[1] = ((const uint4*)x)[i * 2 + 1];
+      if (ADD) {
+        ((uint4*)&vr)[0] = ((const uint4*)res)[i * 2];
+        ((uint4*)&vr)[1] = ((const uint4*)res)[i * 2 + 1];
+      }
+    }
+    int c0 = (int)((i * 8) % C);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int c = c0 + k;
+      float f = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      if (ADD) f += to_f<T>(vr.e[k]);
+      vy.e[k] = from_f<T>(f > 0.f ? f : 0.f);
+    }
+    if (sizeof(T) == 2) {
+      ((uint4*)y)[i] = vy.u;
+    } else {
+      ((uint4*)y)[i * 2] = ((uint4*)&vy)[0];
+      ((uint4*)y)[i * 2 + 1] = ((uint4*)&vy)[1];
+    }
+  }
+}
+
+// ---- backward stats: g = dy*(y>0); per-channel sum(g), sum(g*xhat);
+//      optionally materialize g (residual gradient for the ADD variant)
+template <typename T, bool WRITE_G>
+__global__ __launch_bounds__(256) void bn_bwd_stats_k(
+    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    T* __restrict__ g_out, const float* __restrict__ mean,
+    const float* __restrict__ invstd, long long total, int C,
+    float* __restrict__ sum_g, float* __restrict__ sum_gx) {
+  extern __shared__ float lds[];
+  float* lg = lds;
+  float* lgx = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const long long nvec = total / 8;
+  for (long long i = tid; i < nvec; i += nthreads) {
+    union { uint4 u; T e[8]; } vx, vy, vd, vg;
+    if (sizeof(T) == 2) {
+      vx.u = ((const uint4*)x)[i];
+      vy.u = ((const uint4*)y)[i];
+      vd.u = ((const uint4*)dy)[i];
+    } else {
+      ((uint4*)&vx)[0] = ((const uint4*)x)[i * 2];
+      ((uint4*)&vx)[1] = ((const uint4*)x)[i * 2 + 1];
+      ((uint4*)&vy)[0] = ((const uint4*)y)[i * 2];
+      ((uint4*)&vy)[1] = ((const uint4*)y)[i * 2 + 1];
+      ((uint4*)&vd)[0] = ((const uint4*)dy)[i * 2];
+      ((uint4*)&vd)[1] = ((const uint4*)dy)[i * 2 + 1];
+    }
+    int c0 = (int)((i * 8) % C);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int c = c0 + k;
+      float gv = to_f<T>(vy.e[k]) > 0.f ? to_f<T>(vd.e[k]) : 0.f;
+      float xh = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c];
+      atomicAdd(&lg[c], gv);
+      atomicAdd(&lgx[c], gv * xh);
+      if (WRITE_G) vg.e[k] = from_f<T>(gv);
+    }
+    if (WRITE_G) {
+      if (sizeof(T) == 2) {
+        ((uint4*)g_out)[i] = vg.u;
+      } else {
+        ((uint4*)g_out)[i * 2] = ((uint4*)&vg)[0];
+        ((uint4*)g_out)[i * 2 + 1] = ((uint4*)&vg)[1];
+      }
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lg[c] != 0.f || lgx[c] != 0.f) {
+      atomicAdd(&sum_g[c], lg[c]);
+      atomicAdd(&sum_gx[c], lgx[c]);
+    }
+  }
+}
+
+// ---- backward apply: dx = gamma*invstd*(g - sum_g/n - xhat*sum_gx/n) ------
+template <typename T>
+__global__ __launch_bounds__(256) void bn_bwd_apply_k(
+    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    T* __restrict__ dx, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ sum_g, const float* __restrict__ sum_gx,
+    long long total, int C, float inv_count) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const long long nvec = total / 8;
+  for (long long i = tid; i < nvec; i += nthreads) {
+    union { uint4 u; T e[8]; } vx, vy, vd, vo;
+    if (sizeof(T) == 2) {
+      vx.u = ((const uint4*)x)[i];
+      vy.u = ((const uint4*)y)[i];
+      vd.u = ((const uint4*)dy)[i];
+    } else {
+      ((uint4*)&vx)[0] = ((const uint4*)x)[i * 2];
+      ((uint4*)&vx)[1] = ((const uint4*)x)[i * 2 + 1];
+      ((uint4*)&vy)[0] = ((const uint4*)y)[i * 2];
+      ((uint4*)&vy)[1] = ((const uint4*)y)[i * 2 + 1];
+      ((uint4*)&vd)[0] = ((const uint4*)dy)[i * 2];
+      ((uint4*)&vd)[1] = ((const uint4*)dy)[i * 2 + 1];
+    }
+    int c0 = (int)((i * 8) % C);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int c = c0 + k;
+      float gv = to_f<T>(vy.e[k]) > 0.f ? to_f<T>(vd.e[k]) : 0.f;
+      float xh = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c];
+      float v = gamma[c] * invstd[c] *
+                (gv - sum_g[c] * inv_count - xh * sum_gx[c] * inv_count);
+      vo.e[k] = from_f<T>(v);
+    }
+    if (sizeof(T) == 2) {
+      ((uint4*)dx)[i] = vo.u;
+    } else {
+      ((uint4*)dx)[i * 2] = ((uint4*)&vo)[0];
+      ((uint4*)dx)[i * 2 + 1] = ((uint4*)&vo)[1];
+    }
+  }
+}
+
+int grid_for(long long total) {
+  long long blocks = (total / 8 + 255) / 256;
+  return (int)std::min<long long>(blocks > 0 ? blocks : 1, 2048);
+}
+
+}  // namespace
+
+#define DISPATCH_T(DT, FN)                                        \
+  switch (DT) {                                                   \
+    case DT_F32: { using scalar_t = float; FN; break; }           \
+    case DT_F16: { using scalar_t = __half; FN; break; }          \
+    case DT_BF16: { using scalar_t = __hip_bfloat16; FN; break; } \
+    default: return hipErrorInvalidValue;                         \
+  }
+
+hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
+                         float* sums, float* sqs, hipStream_t stream) {
+  int blocks = grid_for(total);
+  size_t lds = 2 * (size_t)C * sizeof(float);
+  DISPATCH_T(dt, (bn_stats_k<scalar_t><<<blocks, 256, lds, stream>>>(
+                     (const scalar_t*)x, total, C, sums, sqs)));
+  return hipGetLastError();
+}
+
+hipError_t BnApplyReluLaunch(const void* x, const void* res, void* y,
+                             const float* mean, const float* invstd,
+                             const float* gamma, const float* beta,
+                             long long total, int C, int dt,
+                             hipStream_t stream) {
+  int blocks = grid_for(total);
+  if (res) {
+    DISPATCH_T(dt, (bn_apply_relu_k<scalar_t, true><<<blocks, 256, 0, stream>>>(
+                       (const scalar_t*)x, (const scalar_t*)res, (scalar_t*)y,
+                       mean, invstd, gamma, beta, total, C)));
+  } else {
+    DISPATCH_T(dt,
+               (bn_apply_relu_k<scalar_t, false><<<blocks, 256, 0, stream>>>(
+                   (const scalar_t*)x, nullptr, (scalar_t*)y, mean, invstd,
+                   gamma, beta, total, C)));
+  }
+  return hipGetLastError();
+}
+
+hipError_t BnBwdStatsLaunch(const void* x, const void* y, const void* dy,
+                            void* g_out, const float* mean,
+                            const float* invstd, long long total, int C,
+                            int dt, float* sum_g, float* sum_gx,
+                            hipStream_t stream) {
+  int blocks = grid_for(total);
+  size_t lds = 2 * (size_t)C * sizeof(float);
+  if (g_out) {
+    DISPATCH_T(dt,
+               (bn_bwd_stats_k<scalar_t, true><<<blocks, 256, lds, stream>>>(
+                   (const scalar_t*)x, (const scalar_t*)y, (const scalar_t*)dy,
+                   (scalar_t*)g_out, mean, invstd, total, C, sum_g, sum_gx)));
+  } else {
+    DISPATCH_T(dt,
+               (bn_bwd_stats_k<scalar_t, false><<<blocks, 256, lds, stream>>>(
+                   (const scalar_t*)x, (const scalar_t*)y, (const scalar_t*)dy,
+                   nullptr, mean, invstd, total, C, sum_g, sum_gx)));
+  }
+  return hipGetLastError();
+}
+
+hipError_t BnBwdApplyLaunch(const void* x, const void* y, const void* dy,
+                            void* dx, const float* mean, const float* invstd,
+                            const float* gamma, const float* sum_g,
+                            const float* sum_gx, long long total, int C,
+                            int dt, float inv_count, hipStream_t stream) {
+  int blocks = grid_for(total);
+  DISPATCH_T(dt, (bn_bwd_apply_k<scalar_t><<<blocks, 256, 0, stream>>>(
+                     (const scalar_t*)x, (const scalar_t*)y,
+                     (const scalar_t*)dy, (scalar_t*)dx, mean, invstd, gamma,
+                     sum_g, sum_gx, total, C, inv_count)));
+  return hipGetLastError();
+}
+
+#undef DISPATCH_T
+
+}  // namespace gpu
+}  // namespace hvd

@@ -747,6 +747,87 @@ void FusedSgdStep(std::vector<at::Tensor>& params,
   flush();
 }
 
+// ---- Fused BN(+Add)+ReLU host side ----------------------------------------
+// Returns {y, save_mean, save_invstd}.  Small per-channel math ([C] tensors)
+// runs through ATen; the big passes are the CDNA4 kernels.
+std::vector<at::Tensor> FusedBnReluForward(at::Tensor x, at::Tensor residual,
+                                           at::Tensor gamma, at::Tensor beta,
+                                           at::Tensor running_mean,
+                                           at::Tensor running_var,
+                                           double momentum, double eps) {
+  TORCH_CHECK(x.dim() == 4, "fused BN expects NCHW logical 4-D input");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "fused BN requires channels_last layout");
+  int64_t C = x.size(1);
+  TORCH_CHECK(C % 8 == 0 && C <= 4096, "fused BN requires C % 8 == 0, C <= 4096");
+  int device = (int)x.get_device();
+  c10::hip::HIPGuard guard(device);
+  hipStream_t stream = c10::hip::getCurrentHIPStream(device).stream();
+  int64_t total = x.numel();
+  int64_t count = total / C;
+  int dt = (int)DataTypeFromTorch(x.scalar_type());
+
+  auto fopts = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor sums = at::zeros({2, C}, fopts);
+  HIP_CHECK(BnStatsLaunch(x.data_ptr(), total, (int)C, dt,
+                          sums[0].data_ptr<float>(), sums[1].data_ptr<float>(),
+                          stream));
+  at::Tensor mean = sums[0] / (double)count;
+  at::Tensor var = sums[1] / (double)count - mean * mean;
+  at::Tensor invstd = at::rsqrt(var + eps);
+  // running stats are buffers (requires_grad == false): plain in-place math
+  running_mean.mul_(1 - momentum).add_(mean, momentum);
+  double ub = count > 1 ? (double)count / (count - 1) : 1.0;
+  running_var.mul_(1 - momentum).add_(var * ub, momentum);
+  at::Tensor gamma_f = gamma.to(at::kFloat);
+  at::Tensor beta_f = beta.to(at::kFloat);
+  at::Tensor y = at::empty_like(x);
+  HIP_CHECK(BnApplyReluLaunch(
+      x.data_ptr(), residual.defined() ? residual.data_ptr() : nullptr,
+      y.data_ptr(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      gamma_f.data_ptr<float>(), beta_f.data_ptr<float>(), total, (int)C, dt,
+      stream));
+  return {y, mean, invstd};
+}
+
+// Returns {dx, dgamma, dbeta[, dresidual]}.
+std::vector<at::Tensor> FusedBnReluBackward(at::Tensor x, at::Tensor y,
+                                            at::Tensor dy, at::Tensor mean,
+                                            at::Tensor invstd, at::Tensor gamma,
+                                            bool need_residual_grad) {
+  int device = (int)x.get_device();
+  c10::hip::HIPGuard guard(device);
+  hipStream_t stream = c10::hip::getCurrentHIPStream(device).stream();
+  int64_t C = x.size(1);
+  int64_t total = x.numel();
+  int64_t count = total / C;
+  int dt = (int)DataTypeFromTorch(x.scalar_type());
+  if (!dy.is_contiguous(at::MemoryFormat::ChannelsLast))
+    dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+
+  auto fopts = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  at::Tensor sums = at::zeros({2, C}, fopts);
+  at::Tensor dres;
+  if (need_residual_grad) dres = at::empty_like(x);
+  HIP_CHECK(BnBwdStatsLaunch(x.data_ptr(), y.data_ptr(), dy.data_ptr(),
+                             need_residual_grad ? dres.data_ptr() : nullptr,
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             total, (int)C, dt, sums[0].data_ptr<float>(),
+                             sums[1].data_ptr<float>(), stream));
+  at::Tensor gamma_f = gamma.to(at::kFloat);
+  at::Tensor dx = at::empty_like(x);
+  HIP_CHECK(BnBwdApplyLaunch(x.data_ptr(), y.data_ptr(), dy.data_ptr(),
+                             dx.data_ptr(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), gamma_f.data_ptr<float>(),
+                             sums[0].data_ptr<float>(), sums[1].data_ptr<float>(),
+                             total, (int)C, dt, (float)(1.0 / count), stream));
+  at::Tensor dbeta = sums[0].to(gamma.scalar_type());
+  at::Tensor dgamma = sums[1].to(gamma.scalar_type());
+  std::vector<at::Tensor> out{dx, dgamma, dbeta};
+  if (need_residual_grad) out.push_back(dres);
+  return out;
+}
+
 void WaitAllPending() {
   while (true) {
     {

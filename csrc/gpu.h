@@ -42,6 +42,18 @@ bool RcclUsed();
 // collective path uses the same kernels inside ExecuteAdasum).
 void AdasumCombine(std::vector<at::Tensor>& a, std::vector<at::Tensor>& b);
 
+// Fused BatchNorm(+Add)+ReLU (training forward + backward); see
+// bn_kernels.hip.  residual may be undefined.
+std::vector<at::Tensor> FusedBnReluForward(at::Tensor x, at::Tensor residual,
+                                           at::Tensor gamma, at::Tensor beta,
+                                           at::Tensor running_mean,
+                                           at::Tensor running_var,
+                                           double momentum, double eps);
+std::vector<at::Tensor> FusedBnReluBackward(at::Tensor x, at::Tensor y,
+                                            at::Tensor dy, at::Tensor mean,
+                                            at::Tensor invstd, at::Tensor gamma,
+                                            bool need_residual_grad);
+
 // Fused SGD step on torch's current stream (one kernel for all buckets;
 // empty `momenta` = plain SGD).  fp32 tensors.
 void FusedSgdStep(std::vector<at::Tensor>& params,

@@ -70,5 +70,26 @@ hipError_t FusedSgdLaunch(const SgdBatchArgs& args, float lr, float momentum,
                           float weight_decay, float dampening, bool nesterov,
                           hipStream_t stream);
 
+// ---- Fused BatchNorm(+Add)+ReLU (bn_kernels.hip) --------------------------
+// NHWC dense activations [total=N*H*W rows, C channels], C % 8 == 0,
+// dt in {DT_F32, DT_F16, DT_BF16}; stats/params fp32.
+hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
+                         float* sums, float* sqs, hipStream_t stream);
+hipError_t BnApplyReluLaunch(const void* x, const void* res, void* y,
+                             const float* mean, const float* invstd,
+                             const float* gamma, const float* beta,
+                             long long total, int C, int dt,
+                             hipStream_t stream);
+hipError_t BnBwdStatsLaunch(const void* x, const void* y, const void* dy,
+                            void* g_out, const float* mean,
+                            const float* invstd, long long total, int C,
+                            int dt, float* sum_g, float* sum_gx,
+                            hipStream_t stream);
+hipError_t BnBwdApplyLaunch(const void* x, const void* y, const void* dy,
+                            void* dx, const float* mean, const float* invstd,
+                            const float* gamma, const float* sum_g,
+                            const float* sum_gx, long long total, int C,
+                            int dt, float inv_count, hipStream_t stream);
+
 }  // namespace gpu
 }  // namespace hvd

@@ -15,34 +15,47 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, in_planes, planes, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d):
+                 norm_layer=nn.BatchNorm2d, fused_bn=False):
         super().__init__()
+        self.fused_bn = fused_bn
         self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
-        self.bn1 = norm_layer(planes)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = norm_layer(planes)
         self.conv3 = nn.Conv2d(planes, planes * self.expansion, 1, bias=False)
-        self.bn3 = norm_layer(planes * self.expansion)
+        if fused_bn:
+            from horovod_amd.ops import FusedBNAddReLU, FusedBNReLU
+            self.bn1 = FusedBNReLU(planes)
+            self.bn2 = FusedBNReLU(planes)
+            self.bn3 = FusedBNAddReLU(planes * self.expansion)
+        else:
+            self.bn1 = norm_layer(planes)
+            self.bn2 = norm_layer(planes)
+            self.bn3 = norm_layer(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x):
         identity = x
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        if self.fused_bn:
+            out = self.bn1(self.conv1(x))
+            out = self.bn2(self.conv2(out))
+            return self.bn3(self.conv3(out), identity)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
         out += identity
         return self.relu(out)
 
 
 class ResNet(nn.Module):
-    def __init__(self, layers, num_classes=1000, norm_layer=nn.BatchNorm2d):
+    def __init__(self, layers, num_classes=1000, norm_layer=nn.BatchNorm2d,
+                 fused_bn=False):
         super().__init__()
         self._norm_layer = norm_layer
+        self._fused_bn = fused_bn
         self.in_planes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = norm_layer(64)
@@ -78,11 +91,12 @@ class ResNet(nn.Module):
                 norm_layer(planes * Bottleneck.expansion),
             )
         layers = [Bottleneck(self.in_planes, planes, stride, downsample,
-                             norm_layer)]
+                             norm_layer, fused_bn=self._fused_bn)]
         self.in_planes = planes * Bottleneck.expansion
         for _ in range(1, blocks):
             layers.append(Bottleneck(self.in_planes, planes,
-                                     norm_layer=norm_layer))
+                                     norm_layer=norm_layer,
+                                     fused_bn=self._fused_bn))
         return nn.Sequential(*layers)
 
     def forward(self, x):

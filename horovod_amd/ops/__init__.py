@@ -1,1 +1,3 @@
 from horovod_amd.ops.fused_sgd import FusedSGD  # noqa: F401
+from horovod_amd.ops.fused_bn import (FusedBNAddReLU,  # noqa: F401
+                                      FusedBNReLU)

@@ -206,3 +206,84 @@ def test_adasum_kernels_golden_gpu(hvd):
             assert torch.allclose(got.float(), exp.float(), rtol=tol,
                                   atol=tol), \
                 (dtype, (got.float() - exp.float()).abs().max())
+
+
+@requires_gpu
+def test_fused_bn_relu_matches_torch(hvd):
+    """FusedBNReLU / FusedBNAddReLU vs eager BN+ReLU (fwd, bwd, running
+    stats), fp32 and bf16, NHWC."""
+    from horovod_amd.ops import FusedBNAddReLU, FusedBNReLU
+    for dtype, tol in ((torch.float32, 2e-4), (torch.bfloat16, 5e-2)):
+        torch.manual_seed(6)
+        N, C, H, W = 8, 64, 14, 14
+        x = (torch.randn(N, C, H, W, device="cuda")
+             .to(memory_format=torch.channels_last).to(dtype)
+             .requires_grad_(True))
+        xr = x.detach().clone().requires_grad_(True)
+
+        fused = FusedBNReLU(C).cuda().train()
+        ref = torch.nn.BatchNorm2d(C).cuda().train()
+        ref.load_state_dict(fused.state_dict())
+
+        y = fused(x)
+        yr = torch.nn.functional.relu(ref(xr))
+        assert torch.allclose(y.float(), yr.float(), rtol=tol, atol=tol), \
+            (dtype, (y.float() - yr.float()).abs().max())
+        assert torch.allclose(fused.running_mean, ref.running_mean,
+                              atol=1e-4)
+        assert torch.allclose(fused.running_var, ref.running_var, atol=1e-3)
+
+        g = torch.randn_like(y)
+        y.backward(g)
+        yr.backward(g)
+        assert torch.allclose(x.grad.float(), xr.grad.float(), rtol=tol,
+                              atol=tol), \
+            (dtype, (x.grad.float() - xr.grad.float()).abs().max())
+        assert torch.allclose(fused.weight.grad, ref.weight.grad, rtol=1e-2,
+                              atol=1e-2)
+        assert torch.allclose(fused.bias.grad, ref.bias.grad, rtol=1e-2,
+                              atol=1e-2)
+
+        # Add variant
+        a = (torch.randn(N, C, H, W, device="cuda")
+             .to(memory_format=torch.channels_last).to(dtype)
+             .requires_grad_(True))
+        r = (torch.randn(N, C, H, W, device="cuda")
+             .to(memory_format=torch.channels_last).to(dtype)
+             .requires_grad_(True))
+        ar = a.detach().clone().requires_grad_(True)
+        rr = r.detach().clone().requires_grad_(True)
+        fadd = FusedBNAddReLU(C).cuda().train()
+        radd = torch.nn.BatchNorm2d(C).cuda().train()
+        radd.load_state_dict(fadd.state_dict())
+        ya = fadd(a, r)
+        yra = torch.nn.functional.relu(radd(ar) + rr)
+        assert torch.allclose(ya.float(), yra.float(), rtol=tol, atol=tol)
+        ga = torch.randn_like(ya)
+        ya.backward(ga)
+        yra.backward(ga)
+        assert torch.allclose(a.grad.float(), ar.grad.float(), rtol=tol,
+                              atol=tol), \
+            (dtype, (a.grad.float() - ar.grad.float()).abs().max())
+        assert torch.allclose(r.grad.float(), rr.grad.float(), rtol=tol,
+                              atol=tol)
+
+
+@requires_gpu
+def test_fused_bn_resnet_step(hvd):
+    from horovod_amd.models import resnet50
+    torch.manual_seed(3)
+    model = resnet50(fused_bn=True).cuda().to(
+        memory_format=torch.channels_last)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    data = torch.randn(8, 3, 224, 224, device="cuda").to(
+        memory_format=torch.channels_last)
+    target = torch.randint(0, 1000, (8,), device="cuda")
+    for _ in range(2):
+        opt.zero_grad()
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            loss = torch.nn.functional.cross_entropy(model(data), target)
+        loss.backward()
+        opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
