@@ -37,6 +37,10 @@ __device__ __forceinline__ __hip_bfloat16 from_f<__hip_bfloat16>(float v) {
 }
 
 // ---- forward stats: per-channel sum and sum-of-squares --------------------
+// Fixed-channel decomposition: the host sizes the grid so
+// (threads * 8) % C == 0, making every thread revisit the SAME 8 channels
+// each grid-stride iteration.  Partials live in registers; one LDS flush
+// per thread at block end, one global atomicAdd per (block, channel).
 template <typename T>
 __global__ __launch_bounds__(256) void bn_stats_k(const T* __restrict__ x,
                                                   long long total, int C,
@@ -49,8 +53,9 @@ __global__ __launch_bounds__(256) void bn_stats_k(const T* __restrict__ x,
   __syncthreads();
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long nthreads = (long long)gridDim.x * blockDim.x;
-  // 8 contiguous elements per iteration (16 B for bf16/fp16)
   const long long nvec = total / 8;
+  const int c0 = (int)((tid * 8) % C);
+  float rsum[8] = {0}, rsq[8] = {0};
   for (long long i = tid; i < nvec; i += nthreads) {
     union { uint4 u; T e[8]; } v;
     // T = float: 8 floats = 32 B -> two uint4 loads
@@ -60,13 +65,17 @@ __global__ __launch_bounds__(256) void bn_stats_k(const T* __restrict__ x,
       ((uint4*)&v)[0] = ((const uint4*)x)[i * 2];
       ((uint4*)&v)[1] = ((const uint4*)x)[i * 2 + 1];
     }
-    int c0 = (int)((i * 8) % C);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
       float f = to_f<T>(v.e[k]);
-      atomicAdd(&lsum[c0 + k], f);
-      atomicAdd(&lsq[c0 + k], f * f);
+      rsum[k] += f;
+      rsq[k] += f * f;
     }
+  }
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    atomicAdd(&lsum[c0 + k], rsum[k]);
+    atomicAdd(&lsq[c0 + k], rsq[k]);
   }
   __syncthreads();
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -87,6 +96,17 @@ __global__ __launch_bounds__(256) void bn_apply_relu_k(
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long nthreads = (long long)gridDim.x * blockDim.x;
   const long long nvec = total / 8;
+  // fixed-channel decomposition (see bn_stats_k): per-thread channel params
+  // load once into registers
+  const int c0 = (int)((tid * 8) % C);
+  float rm[8], ri[8], rg[8], rb[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    rm[k] = mean[c0 + k];
+    ri[k] = invstd[c0 + k];
+    rg[k] = gamma[c0 + k];
+    rb[k] = beta[c0 + k];
+  }
   for (long long i = tid; i < nvec; i += nthreads) {
     union { uint4 u; T e[8]; } vx, vr, vy;
     if (sizeof(T) == 2) {
@@ -100,11 +120,9 @@ __global__ __launch_bounds__(256) void bn_apply_relu_k(
         ((uint4*)&vr)[1] = ((const uint4*)res)[i * 2 + 1];
       }
     }
-    int c0 = (int)((i * 8) % C);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      int c = c0 + k;
-      float f = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      float f = (to_f<T>(vx.e[k]) - rm[k]) * ri[k] * rg[k] + rb[k];
       if (ADD) f += to_f<T>(vr.e[k]);
       vy.e[k] = from_f<T>(f > 0.f ? f : 0.f);
     }
@@ -133,6 +151,13 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_k(
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long nthreads = (long long)gridDim.x * blockDim.x;
   const long long nvec = total / 8;
+  const int c0 = (int)((tid * 8) % C);
+  float rmean[8], rinv[8], racc_g[8] = {0}, racc_gx[8] = {0};
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    rmean[k] = mean[c0 + k];
+    rinv[k] = invstd[c0 + k];
+  }
   for (long long i = tid; i < nvec; i += nthreads) {
     union { uint4 u; T e[8]; } vx, vy, vd, vg;
     if (sizeof(T) == 2) {
@@ -147,14 +172,12 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_k(
       ((uint4*)&vd)[0] = ((const uint4*)dy)[i * 2];
       ((uint4*)&vd)[1] = ((const uint4*)dy)[i * 2 + 1];
     }
-    int c0 = (int)((i * 8) % C);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      int c = c0 + k;
       float gv = to_f<T>(vy.e[k]) > 0.f ? to_f<T>(vd.e[k]) : 0.f;
-      float xh = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c];
-      atomicAdd(&lg[c], gv);
-      atomicAdd(&lgx[c], gv * xh);
+      float xh = (to_f<T>(vx.e[k]) - rmean[k]) * rinv[k];
+      racc_g[k] += gv;
+      racc_gx[k] += gv * xh;
       if (WRITE_G) vg.e[k] = from_f<T>(gv);
     }
     if (WRITE_G) {
@@ -165,6 +188,11 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_k(
         ((uint4*)g_out)[i * 2 + 1] = ((uint4*)&vg)[1];
       }
     }
+  }
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    atomicAdd(&lg[c0 + k], racc_g[k]);
+    atomicAdd(&lgx[c0 + k], racc_gx[k]);
   }
   __syncthreads();
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -186,6 +214,16 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_k(
   const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long nthreads = (long long)gridDim.x * blockDim.x;
   const long long nvec = total / 8;
+  const int c0 = (int)((tid * 8) % C);
+  float rmean[8], rinv[8], rgi[8], rsg[8], rsgx[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    rmean[k] = mean[c0 + k];
+    rinv[k] = invstd[c0 + k];
+    rgi[k] = gamma[c0 + k] * rinv[k];
+    rsg[k] = sum_g[c0 + k] * inv_count;
+    rsgx[k] = sum_gx[c0 + k] * inv_count;
+  }
   for (long long i = tid; i < nvec; i += nthreads) {
     union { uint4 u; T e[8]; } vx, vy, vd, vo;
     if (sizeof(T) == 2) {
@@ -200,14 +238,11 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_k(
       ((uint4*)&vd)[0] = ((const uint4*)dy)[i * 2];
       ((uint4*)&vd)[1] = ((const uint4*)dy)[i * 2 + 1];
     }
-    int c0 = (int)((i * 8) % C);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      int c = c0 + k;
       float gv = to_f<T>(vy.e[k]) > 0.f ? to_f<T>(vd.e[k]) : 0.f;
-      float xh = (to_f<T>(vx.e[k]) - mean[c]) * invstd[c];
-      float v = gamma[c] * invstd[c] *
-                (gv - sum_g[c] * inv_count - xh * sum_gx[c] * inv_count);
+      float xh = (to_f<T>(vx.e[k]) - rmean[k]) * rinv[k];
+      float v = rgi[k] * (gv - rsg[k] - xh * rsgx[k]);
       vo.e[k] = from_f<T>(v);
     }
     if (sizeof(T) == 2) {
@@ -219,9 +254,17 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_k(
   }
 }
 
-int grid_for(long long total) {
+int grid_for(long long total, int C) {
   long long blocks = (total / 8 + 255) / 256;
-  return (int)std::min<long long>(blocks > 0 ? blocks : 1, 2048);
+  blocks = std::min<long long>(blocks > 0 ? blocks : 1, 2048);
+  // fixed-channel decomposition invariant: (blocks*256*8) % C == 0 so each
+  // thread's channel octet is stride-invariant.  vecs_per_row = C/8 <= 512;
+  // round blocks up to a multiple of ceil(C/8/256) and ensure divisibility.
+  long long vpr = C / 8;  // vectors per row
+  long long g = std::__gcd((long long)256, vpr);
+  long long mult = vpr / g;  // blocks must be a multiple of this
+  blocks = ((blocks + mult - 1) / mult) * mult;
+  return (int)blocks;
 }
 
 }  // namespace
@@ -236,7 +279,7 @@ int grid_for(long long total) {
 
 hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
                          float* sums, float* sqs, hipStream_t stream) {
-  int blocks = grid_for(total);
+  int blocks = grid_for(total, C);
   size_t lds = 2 * (size_t)C * sizeof(float);
   DISPATCH_T(dt, (bn_stats_k<scalar_t><<<blocks, 256, lds, stream>>>(
                      (const scalar_t*)x, total, C, sums, sqs)));
@@ -248,7 +291,7 @@ hipError_t BnApplyReluLaunch(const void* x, const void* res, void* y,
                              const float* gamma, const float* beta,
                              long long total, int C, int dt,
                              hipStream_t stream) {
-  int blocks = grid_for(total);
+  int blocks = grid_for(total, C);
   if (res) {
     DISPATCH_T(dt, (bn_apply_relu_k<scalar_t, true><<<blocks, 256, 0, stream>>>(
                        (const scalar_t*)x, (const scalar_t*)res, (scalar_t*)y,
@@ -267,7 +310,7 @@ hipError_t BnBwdStatsLaunch(const void* x, const void* y, const void* dy,
                             const float* invstd, long long total, int C,
                             int dt, float* sum_g, float* sum_gx,
                             hipStream_t stream) {
-  int blocks = grid_for(total);
+  int blocks = grid_for(total, C);
   size_t lds = 2 * (size_t)C * sizeof(float);
   if (g_out) {
     DISPATCH_T(dt,
@@ -288,7 +331,7 @@ hipError_t BnBwdApplyLaunch(const void* x, const void* y, const void* dy,
                             const float* gamma, const float* sum_g,
                             const float* sum_gx, long long total, int C,
                             int dt, float inv_count, hipStream_t stream) {
-  int blocks = grid_for(total);
+  int blocks = grid_for(total, C);
   DISPATCH_T(dt, (bn_bwd_apply_k<scalar_t><<<blocks, 256, 0, stream>>>(
                      (const scalar_t*)x, (const scalar_t*)y,
                      (const scalar_t*)dy, (scalar_t*)dx, mean, invstd, gamma,
