@@ -42,7 +42,9 @@ def parse_args():
                    help="use the CDNA4 fused SGD step kernel (default)")
     p.add_argument("--no-fused-sgd", dest="fused_sgd", action="store_false")
     p.add_argument("--fused-bn", action="store_true",
-                   help="use the fused CDNA4 BatchNorm(+Add)+ReLU kernels")
+                   help="use the fused CDNA4 BatchNorm(+Add)+ReLU kernels "
+                        "(experimental: numerics-verified, currently slower "
+                        "than MIOpen's BN at batch 64)")
     p.add_argument("--persistent-grads", action="store_true",
                    help="zero_grad(set_to_none=False): keep gradient buffers "
                         "allocated across steps")
