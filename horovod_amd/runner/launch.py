@@ -136,6 +136,9 @@ def build_parser():
                         help="elastic: maximum np")
     parser.add_argument("--host-discovery-script", default=None,
                         help="elastic: executable printing host:slots lines")
+    parser.add_argument("-cb", "--check-build", action="store_true",
+                        help="print available features/frameworks and exit "
+                             "(reference: horovodrun --check-build)")
     parser.add_argument("--config-file", default=None,
                         help="YAML file with launcher settings (reference: "
                              "launch.py:581-585)")
@@ -157,9 +160,30 @@ def apply_config_file(args):
     return args
 
 
+def check_build():
+    print("""horovod_amd (MI355X-native)
+
+Available frameworks:
+    [X] PyTorch (ROCm)
+    [ ] TensorFlow (out of scope: single-stack design)
+    [ ] MXNet (out of scope)
+
+Available controllers:
+    [X] TCP star (gloo-equivalent; built in)
+    [ ] MPI (replaced by design)
+
+Available tensor operations:
+    [X] RCCL over xGMI (GPU)
+    [X] TCP star (CPU)
+    [X] CDNA4 fused kernels: pack/scale/convert, Adasum, SGD, BN+ReLU""")
+    return 0
+
+
 def main(argv=None):
     args = build_parser().parse_args(argv)
     args = apply_config_file(args)
+    if args.check_build:
+        return check_build()
     if not args.command:
         print("hvdrun: no command given", file=sys.stderr)
         return 1

@@ -100,3 +100,10 @@ def test_spark_ray_graceful_without_deps():
     ex = hvd_ray.RayExecutor(num_workers=2)
     with pytest.raises(ImportError):
         ex.start()
+
+
+def test_check_build(capsys):
+    from horovod_amd.runner.launch import main
+    assert main(["--check-build"]) == 0
+    out = capsys.readouterr().out
+    assert "PyTorch (ROCm)" in out and "RCCL" in out
