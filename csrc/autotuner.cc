@@ -34,6 +34,8 @@ Autotuner::Autotuner(int64_t fusion_bytes, double cycle_time_ms,
     : log_path_(std::move(log_path)) {
   current_ = {fusion_bytes, cycle_time_ms};
   best_ = current_;
+  if (const char* w = std::getenv("HOROVOD_AUTOTUNE_WINDOW_SECONDS"))
+    window_sec_ = atof(w);
   // coarse warm-start grid (reference seeds with a grid before the GP takes
   // over, parameter_manager.cc:44-61)
   for (double mb : {8.0, 32.0, 64.0, 128.0}) {

@@ -359,3 +359,83 @@ def test_gradient_predivide_np2():
         for p, r, e in zip(model.parameters(), ref, expected):
             assert torch.allclose(p.detach(), r - 0.1 * e, atol=1e-6)
     """)
+
+
+def test_autotune_propagates_np2(tmp_path):
+    """With a short sampling window, the coordinator must propose and
+    propagate new fusion parameters via TUNE responses."""
+    log = str(tmp_path / "at.csv")
+    run_workers(2, """
+        import time
+        import horovod_amd._core as core
+        start = (core.get_fusion_threshold(), core.get_cycle_time_ms())
+        changed = False
+        for i in range(4000):
+            hvd.allreduce(torch.ones(8192), average=False, name="atp")
+            if (core.get_fusion_threshold(),
+                    core.get_cycle_time_ms()) != start:
+                changed = True
+                break
+        assert changed, "no TUNE response arrived"
+    """, extra_env={"HOROVOD_AUTOTUNE": "1",
+                    "HOROVOD_AUTOTUNE_WINDOW_SECONDS": "0.3",
+                    "HOROVOD_AUTOTUNE_LOG": log}, timeout=300)
+    assert "," in open(log).read()
+
+
+def test_fuzz_collective_sequences_np2():
+    """Randomized (rank-identical) sequences of mixed collectives; results
+    checked against locally computable expectations."""
+    run_workers(2, """
+        import random
+        rng = random.Random(1234)        # identical schedule on both ranks
+        for i in range(60):
+            kind = rng.choice(["allreduce", "allgather", "broadcast",
+                               "reducescatter", "barrier", "grouped"])
+            n = rng.randint(1, 300)
+            base = torch.arange(n).float()
+            mine = base * (rank + 1)
+            if kind == "allreduce":
+                out = hvd.allreduce(mine, average=False, name=f"f{i}")
+                assert torch.allclose(out, base * 3), (i, kind)
+            elif kind == "allgather":
+                out = hvd.allgather(mine, name=f"f{i}")
+                assert torch.allclose(out, torch.cat([base, base * 2])), i
+            elif kind == "broadcast":
+                root = rng.randint(0, 1)
+                out = hvd.broadcast(mine, root_rank=root, name=f"f{i}")
+                assert torch.allclose(out, base * (root + 1)), i
+            elif kind == "reducescatter":
+                out = hvd.reducescatter(mine, op=hvd.Sum, name=f"f{i}")
+                lo = 0 if rank == 0 else (n + 1) // 2
+                hi = (n + 1) // 2 if rank == 0 else n
+                assert torch.allclose(out, base[lo:hi] * 3), i
+            elif kind == "barrier":
+                hvd.barrier()
+            else:
+                outs = hvd.grouped_allreduce([mine, mine + 1], average=False,
+                                             name=f"f{i}")
+                assert torch.allclose(outs[1], base * 3 + 2), i
+    """, timeout=300)
+
+
+def test_multithreaded_enqueue_np2():
+    """Two python threads enqueue DISTINCT named ops concurrently; the queue
+    and handle manager must be thread-safe.  (Cross-rank op ORDER still must
+    match, so each thread owns its own name space.)"""
+    run_workers(2, """
+        import threading
+        errors = []
+        def worker(tag):
+            try:
+                for i in range(30):
+                    out = hvd.allreduce(torch.ones(64), average=False,
+                                        name=f"mt.{tag}.{i}")
+                    assert out.sum().item() == 128.0
+            except Exception as e:
+                errors.append(e)
+        ts = [threading.Thread(target=worker, args=(t,)) for t in "ab"]
+        for t in ts: t.start()
+        for t in ts: t.join()
+        assert not errors, errors
+    """, timeout=300)
