@@ -4,6 +4,8 @@
 #include <cstdio>
 #include <sstream>
 
+#include "logging.h"
+
 namespace hvd {
 
 namespace {
@@ -712,10 +714,8 @@ void Controller::CheckForStalledTensors() {
     std::ostringstream missing;
     for (size_t i = 0; i < pt.have.size(); ++i)
       if (!pt.have[i]) missing << set.ranks[i] << " ";
-    std::fprintf(stderr,
-                 "[horovod_amd] WARNING: tensor %s stalled for %.0fs; waiting on "
-                 "ranks: %s\n",
-                 kv.first.c_str(), age, missing.str().c_str());
+    HVD_LOG(WARNING, "tensor %s stalled for %.0fs; waiting on ranks: %s",
+            kv.first.c_str(), age, missing.str().c_str());
     if (cfg_.stall_shutdown_sec > 0 && age > cfg_.stall_shutdown_sec) {
       // reference: HOROVOD_STALL_SHUTDOWN_TIME_SECONDS aborts the job when a
       // stall persists (stall_inspector.h:30-97)

@@ -9,6 +9,7 @@
 #include <stdexcept>
 
 #include "gpu.h"
+#include "logging.h"
 #include "timeline.h"
 
 namespace hvd {
@@ -604,7 +605,7 @@ void PerformOperation(GlobalState& st, Response& resp) {
 void Abort(GlobalState& st, const std::string& why) {
   st.aborted = true;
   st.abort_reason = why;
-  std::fprintf(stderr, "[horovod_amd] background loop aborted: %s\n", why.c_str());
+  HVD_LOG(ERROR, "background loop aborted: %s", why.c_str());
   auto s = Status::Aborted(why);
   st.queue.FailAll(s);
   st.handles.FailAll(s);
@@ -706,6 +707,8 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
   }
   st.bg_thread = std::thread([&st] { BackgroundLoop(st); });
   st.initialized = true;
+  HVD_LOG(INFO, "initialized: rank %d/%d (local %d/%d), controller %s:%d",
+          rank, size, local_rank, local_size, addr.c_str(), port);
 }
 
 void ShutdownHorovod() {

@@ -19,6 +19,7 @@
 
 #include "core.h"
 #include "kernels.h"
+#include "logging.h"
 #include "timeline.h"
 
 namespace hvd {
@@ -230,6 +231,8 @@ ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
                               set.local_index(st.rank)));
   ctx.comms[set_id] = comm;
   g_rccl_used = true;
+  HVD_LOG(INFO, "RCCL comm ready: process set %d, %d ranks, device %d",
+          (int)set_id, (int)set.ranks.size(), ctx.device);
   return comm;
 }
 

@@ -130,6 +130,12 @@ def build_parser():
     parser.add_argument("--cache-capacity", type=int, default=None)
     parser.add_argument("--timeline-filename", default=None)
     parser.add_argument("--autotune", action="store_true")
+    parser.add_argument("--network-interface", default=None,
+                        help="NIC(s) for RCCL bootstrap "
+                             "(sets NCCL_SOCKET_IFNAME)")
+    parser.add_argument("--log-level", default=None,
+                        choices=["trace", "debug", "info", "warning",
+                                 "error", "fatal"])
     parser.add_argument("--min-np", type=int, default=None,
                         help="elastic: minimum np")
     parser.add_argument("--max-np", type=int, default=None,
@@ -202,6 +208,10 @@ def main(argv=None):
         env["HOROVOD_TIMELINE"] = args.timeline_filename
     if args.autotune:
         env["HOROVOD_AUTOTUNE"] = "1"
+    if args.network_interface:
+        env["NCCL_SOCKET_IFNAME"] = args.network_interface
+    if args.log_level:
+        env["HOROVOD_LOG_LEVEL"] = args.log_level
 
     hosts = args.hosts
     if args.hostfile:
