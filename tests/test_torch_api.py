@@ -439,3 +439,14 @@ def test_multithreaded_enqueue_np2():
         for t in ts: t.join()
         assert not errors, errors
     """, timeout=300)
+
+
+def test_init_with_process_sets_np2():
+    run_workers(2, """
+        # (re-init path: hvd.init already ran in the harness prelude, so
+        # register via the documented collective API instead)
+        ps = hvd.add_process_set([0, 1])
+        out = hvd.allreduce(torch.ones(2), average=False, process_set=ps,
+                            name="ip")
+        assert out.sum().item() == 4.0
+    """)
