@@ -4,7 +4,9 @@
 #include <cstdio>
 #include <sstream>
 
+#include "core.h"
 #include "logging.h"
+#include "timeline.h"
 
 namespace hvd {
 
@@ -202,9 +204,11 @@ void Controller::RemoveProcessSet(int32_t id) { process_sets_.erase(id); }
 ResponseList Controller::RunCycle(std::vector<Request> new_requests,
                                   bool shutdown_requested) {
   // -- 1. classify ----------------------------------------------------------
+  auto tl = GetTimeline(State());
   std::vector<Request> slow;  // to coordinator this cycle
   std::vector<int> my_invalid_slots;
   for (auto& req : new_requests) {
+    if (tl) tl->NegotiateStart(req.name);
     if (req.type == RequestType::JOIN || req.type == RequestType::BARRIER ||
         req.type == RequestType::ALLTOALL) {
       // never cached: alltoall carries runtime splits; join/barrier are

@@ -563,7 +563,10 @@ void PerformOperation(GlobalState& st, Response& resp) {
   }
 
   auto tl = GetTimeline(st);
-  if (tl) tl->OpStart(resp);
+  if (tl) {
+    for (auto& n : resp.names) tl->NegotiateEnd(n);
+    tl->OpStart(resp);
+  }
 
   // Response device is the normalized marker (-1 CPU, -2 GPU) — the GPU path
   // must run on every rank (even relay-only non-members) so the RCCL comm

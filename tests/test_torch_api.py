@@ -257,6 +257,7 @@ def test_timeline_np2(tmp_path):
     data = _json.load(open(tl))
     names = [e.get("name") for e in data]
     assert "ALLREDUCE" in names, names[:10]
+    assert "NEGOTIATE" in names, "per-tensor negotiate phases missing"
     data1 = _json.load(open(tl + ".1"))
     assert data1
 
