@@ -451,3 +451,16 @@ def test_init_with_process_sets_np2():
                             name="ip")
         assert out.sum().item() == 4.0
     """)
+
+
+def test_metric_averaging_np2():
+    run_workers(2, """
+        from horovod_amd.torch import MetricAverager, avg_metrics
+        out = avg_metrics({"loss": 1.0 + rank, "acc": 0.5 * (rank + 1)})
+        assert abs(out["loss"] - 1.5) < 1e-9 and abs(out["acc"] - 0.75) < 1e-9
+        m = MetricAverager()
+        m.update("loss", 2.0 + rank, n=rank + 1)  # weighted
+        avg = m.averages()
+        # (2.0*1 + 3.0*2) / 3 = 8/3
+        assert abs(avg["loss"] - 8.0 / 3) < 1e-9, avg
+    """)
