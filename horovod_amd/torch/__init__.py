@@ -30,6 +30,7 @@ from horovod_amd.torch.mpi_ops import (add_process_set,  # noqa: F401
 from horovod_amd.common.process_sets import (ProcessSet,  # noqa: F401
                                              global_process_set)
 from horovod_amd.common.exceptions import HorovodInternalError  # noqa: F401
+from horovod_amd.torch.lr_scheduler import WarmupScheduler  # noqa: F401
 from horovod_amd.torch.metrics import MetricAverager, avg_metrics  # noqa: F401
 from horovod_amd.torch.optimizer import DistributedOptimizer  # noqa: F401
 from horovod_amd.torch.sync_batch_norm import SyncBatchNorm  # noqa: F401

@@ -464,3 +464,21 @@ def test_metric_averaging_np2():
         # (2.0*1 + 3.0*2) / 3 = 8/3
         assert abs(avg["loss"] - 8.0 / 3) < 1e-9, avg
     """)
+
+
+def test_warmup_scheduler_np2():
+    run_workers(2, """
+        from horovod_amd.torch import WarmupScheduler
+        model = torch.nn.Linear(2, 1)
+        opt = torch.optim.SGD(model.parameters(), lr=0.8)
+        sched = WarmupScheduler(opt, warmup_steps=4)
+        lrs = []
+        for _ in range(6):
+            lrs.append(opt.param_groups[0]["lr"])
+            opt.step()
+            sched.step()
+        # starts at lr/size = 0.4, ramps to 0.8, stays
+        assert abs(lrs[0] - 0.4) < 1e-9, lrs
+        assert abs(lrs[-1] - 0.8) < 1e-9, lrs
+        assert lrs == sorted(lrs), lrs
+    """)
