@@ -482,3 +482,41 @@ def test_warmup_scheduler_np2():
         assert abs(lrs[-1] - 0.8) < 1e-9, lrs
         assert lrs == sorted(lrs), lrs
     """)
+
+
+def test_adasum_golden_np4():
+    rng = np.random.RandomState(11)
+    vecs = [rng.randn(48).astype(np.float32) for _ in range(4)]
+    expected = _adasum_golden(vecs)
+    run_workers(4, f"""
+        import numpy as np
+        allv = {[v.tolist() for v in vecs]!r}
+        t = torch.tensor(allv[rank], dtype=torch.float32)
+        out = hvd.allreduce(t, op=hvd.Adasum, name="adasum4")
+        expected = np.array({expected.tolist()!r})
+        assert np.allclose(out.numpy(), expected, atol=1e-5), \\
+            np.abs(out.numpy() - expected).max()
+    """)
+
+
+def test_optimizer_groups_with_compression_np2():
+    run_workers(2, """
+        from horovod_amd.torch.compression import Compression
+        torch.manual_seed(13)
+        model = torch.nn.Sequential(torch.nn.Linear(8, 16),
+                                    torch.nn.ReLU(), torch.nn.Linear(16, 4))
+        opt = hvd.DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.05),
+            named_parameters=model.named_parameters(), groups=2,
+            compression=Compression.fp16)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(40 + rank)
+        for _ in range(3):
+            opt.zero_grad()
+            model(torch.randn(4, 8)).sum().backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        g = hvd.allgather(flat.unsqueeze(0), name="gc")
+        assert torch.allclose(g[0], g[1], atol=1e-5), \
+            (g[0] - g[1]).abs().max()
+    """)
