@@ -84,3 +84,20 @@ def test_bert_cpu_np2():
         g = hvd.allgather(flat[:500].unsqueeze(0), name="bert_p")
         assert torch.allclose(g[0], g[1], atol=1e-6)
     """, timeout=300)
+
+
+def test_elastic_cli_example(tmp_path):
+    """`hvdrun --host-discovery-script` end-to-end with the shipped elastic
+    MNIST example (reference: test_elastic_torch.py integration model)."""
+    script = tmp_path / "discover.sh"
+    script.write_text("#!/bin/sh\necho '127.0.0.1:2'\n")
+    script.chmod(0o755)
+    r = subprocess.run(
+        [sys.executable, "-m", "horovod_amd.runner.launch",
+         "--host-discovery-script", str(script), "--min-np", "1",
+         "--max-np", "2",
+         sys.executable, os.path.join(REPO, "examples",
+                                      "pytorch_elastic_mnist.py"),
+         "--epochs", "1", "--samples", "256"],
+        env=_env(), capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
