@@ -135,19 +135,20 @@ def _resolve_scales(op, average, prescale_factor, postscale_factor, process_set)
 
 
 class _HandleInfo:
-    __slots__ = ("native", "outputs_n", "kind")
+    __slots__ = ("native", "outputs_n", "kind", "post_divisor")
 
-    def __init__(self, native, outputs_n=1, kind="op"):
+    def __init__(self, native, outputs_n=1, kind="op", post_divisor=None):
         self.native = native
         self.outputs_n = outputs_n
         self.kind = kind
+        self.post_divisor = post_divisor
 
 
 _handles = {}
 
 
-def _register(native_handle, outputs_n=1, kind="op"):
-    info = _HandleInfo(native_handle, outputs_n, kind)
+def _register(native_handle, outputs_n=1, kind="op", post_divisor=None):
+    info = _HandleInfo(native_handle, outputs_n, kind, post_divisor)
     _handles[native_handle] = info
     return native_handle
 
@@ -162,6 +163,12 @@ def synchronize(handle):
     info = _handles.pop(handle, None)
     _core.flush()  # cut the cycle-pacing window: we are about to block
     outs, extra, result_int = _translate_error(_core.wait, handle)
+    if info is not None and info.post_divisor:
+        # integer Average: floor-divide the summed result (reference:
+        # DivideInPlace, mpi_ops_v2.cc:62-68)
+        for o in outs:
+            if o is not None and o.numel():
+                o.floor_divide_(info.post_divisor)
     if info is not None and info.kind == "join":
         return result_int
     if info is not None and info.kind == "alltoall_splits":
@@ -179,11 +186,11 @@ def wait(handle):
 # allreduce
 # ---------------------------------------------------------------------------
 def _allreduce_async_impl(tensors, outputs, name, true_op, pre, post, ps_id,
-                          wire_code, kind="op"):
+                          wire_code, kind="op", post_divisor=None):
     names = list(name) if isinstance(name, (list, tuple)) else [name]
     h = _core.allreduce_async(tensors, outputs, names, true_op, pre, post,
                               ps_id, wire_code)
-    return _register(h, len(tensors), kind)
+    return _register(h, len(tensors), kind, post_divisor)
 
 
 def allreduce_async(tensor, average=None, name=None, op=None,
@@ -212,12 +219,19 @@ def _do_allreduce_async(tensor, output, average, name, op, prescale_factor,
         output = torch.empty_like(tensor)
     true_op, _, pre, post = _resolve_scales(op, average, prescale_factor,
                                             postscale_factor, process_set)
+    post_div = None
+    if not tensor.dtype.is_floating_point and post != 1.0:
+        # integer Average: sum on the wire, floor-divide after (reference
+        # semantics; scale kernels are float-family only)
+        post_div = round(1.0 / post)
+        post = 1.0
     ps_id = _set_id(process_set)
     name = name or _next_name("allreduce")
     wire_code = (_dtype_code(wire_dtype) if wire_dtype is not None
                  else _dtype_code(tensor.dtype))
     return _allreduce_async_impl([tensor], [output], ["allreduce." + name],
-                                 true_op, pre, post, ps_id, wire_code)
+                                 true_op, pre, post, ps_id, wire_code,
+                                 post_divisor=post_div)
 
 
 class HorovodAllreduce(torch.autograd.Function):
@@ -302,13 +316,18 @@ def _grouped_allreduce_impl(tensors, outputs, average, name, op,
             tensors[i] = t.contiguous()
     true_op, _, pre, post = _resolve_scales(op, average, prescale_factor,
                                             postscale_factor, process_set)
+    post_div = None
+    if not tensors[0].dtype.is_floating_point and post != 1.0:
+        post_div = round(1.0 / post)
+        post = 1.0
     ps_id = _set_id(process_set)
     base = name or _next_name("grouped_allreduce")
     names = [f"allreduce.{base}.{i}" for i in range(len(tensors))]
     wire_code = (_dtype_code(wire_dtype) if wire_dtype is not None
                  else _dtype_code(tensors[0].dtype))
     return _allreduce_async_impl(list(tensors), list(outputs), names, true_op,
-                                 pre, post, ps_id, wire_code, kind="grouped")
+                                 pre, post, ps_id, wire_code, kind="grouped",
+                                 post_divisor=post_div)
 
 
 def grouped_allreduce(tensors, average=None, name=None, op=None,

@@ -293,3 +293,14 @@ def test_fused_bn_resnet_step(hvd):
         opt.step()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+@requires_gpu
+def test_integer_average_gpu(hvd):
+    t = torch.tensor([3, -3, 7], dtype=torch.int64, device="cuda")
+    out = hvd.allreduce(t, average=True, name="giavg")
+    assert out.tolist() == [3, -3, 7], out  # n=1: sum==self, divisor 1
+    out2 = hvd.allreduce(torch.arange(64, device="cuda", dtype=torch.int32),
+                         average=True, name="giavg2")
+    assert torch.equal(out2, torch.arange(64, device="cuda",
+                                          dtype=torch.int32))

@@ -448,3 +448,14 @@ def test_join_with_allgather_np2():
             assert out.sum().item() == 4.0  # peer contributes zeros
         hvd.join()
     """, timeout=120)
+
+
+def test_integer_average_np2():
+    """Integer Average floor-divides the sum (reference: DivideInPlace,
+    mpi_ops_v2.cc:62-68) — including negative values."""
+    run_workers(2, """
+        t = torch.tensor([1, -1, 5, -5], dtype=torch.int64) * (rank + 1)
+        out = hvd.allreduce(t, average=True, name="iavg")
+        # sums: [3,-3,15,-15]; floor-div 2 -> [1,-2,7,-8]
+        assert out.tolist() == [1, -2, 7, -8], out
+    """)
