@@ -527,10 +527,14 @@ def reducescatter_async(tensor, op=None, name=None, prescale_factor=1.0,
                         postscale_factor=1.0, process_set=global_process_set):
     true_op, _, pre, post = _resolve_scales(op, None, prescale_factor,
                                             postscale_factor, process_set)
+    post_div = None
+    if not tensor.dtype.is_floating_point and post != 1.0:
+        post_div = round(1.0 / post)  # integer Average: floor-divide post-op
+        post = 1.0
     name = name or _next_name("reducescatter")
     h = _core.reducescatter_async(tensor.contiguous(), "reducescatter." + name,
                                   true_op, pre, post, _set_id(process_set))
-    return _register(h)
+    return _register(h, post_divisor=post_div)
 
 
 class HorovodReducescatter(torch.autograd.Function):

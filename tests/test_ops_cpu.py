@@ -459,3 +459,13 @@ def test_integer_average_np2():
         # sums: [3,-3,15,-15]; floor-div 2 -> [1,-2,7,-8]
         assert out.tolist() == [1, -2, 7, -8], out
     """)
+
+
+def test_reducescatter_integer_average_np2():
+    run_workers(2, """
+        t = torch.tensor([1, -1, 5, -5], dtype=torch.int64) * (rank + 1)
+        out = hvd.reducescatter(t, name="rsiavg")  # default Average
+        full = [1, -2, 7, -8]  # floor(sum/2)
+        mine = full[:2] if rank == 0 else full[2:]
+        assert out.tolist() == mine, (out, mine)
+    """)
