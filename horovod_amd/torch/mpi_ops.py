@@ -259,10 +259,24 @@ class HorovodAllreduce(torch.autograd.Function):
 def allreduce(tensor, average=None, name=None, compression=None, op=None,
               prescale_factor=1.0, postscale_factor=1.0,
               process_set=global_process_set):
-    """Differentiable allreduce; returns a new reduced tensor.
+    """Reduce `tensor` across all processes of the set; differentiable.
 
-    `compression` (hvd.Compression.fp16 / .bf16) maps to wire-dtype
-    conversion inside the fusion pack kernel — no extra python-side pass.
+    Reference semantics: horovod/torch/mpi_ops.py allreduce.
+
+    Arguments:
+        tensor: torch tensor (CPU or MI355X GPU); any dense layout.
+        average: deprecated alias — True => op=Average, False => op=Sum.
+        name: negotiation key.  Ops are matched across ranks by name; pass a
+            stable unique name for anything called concurrently.
+        compression: hvd.Compression.none/fp16/bf16 — wire dtype, converted
+            inside the CDNA4 pack kernel (no python-side copies).
+        op: hvd.Average (default) | Sum | Adasum | Min | Max | Product.
+        prescale_factor/postscale_factor: scalar factors fused into the
+            pack/unpack kernels.  Integer tensors floor-divide after the sum.
+        process_set: subgroup to reduce over.
+
+    Returns a new tensor with the reduction across the set (gradients flow
+    through another allreduce of the same op).
     """
     from horovod_amd.torch.compression import Compression
     wire = None
@@ -406,6 +420,9 @@ class HorovodAllgather(torch.autograd.Function):
 
 
 def allgather(tensor, name=None, process_set=global_process_set):
+    """Concatenate `tensor` from all set members along dim 0; differentiable.
+    First dimensions may differ per rank (variable-gather); other dims must
+    match.  Reference: horovod/torch/mpi_ops.py allgather."""
     return HorovodAllgather.apply(tensor, name, process_set)
 
 
@@ -462,6 +479,9 @@ class HorovodBroadcast(torch.autograd.Function):
 
 
 def broadcast(tensor, root_rank, name=None, process_set=global_process_set):
+    """Return `tensor` broadcast from global rank `root_rank` to every set
+    member; differentiable (grad reduces back to the root).
+    Reference: horovod/torch/mpi_ops.py broadcast."""
     return HorovodBroadcast.apply(tensor, root_rank, name, process_set)
 
 
@@ -512,8 +532,12 @@ class HorovodAlltoall(torch.autograd.Function):
 
 
 def alltoall(tensor, splits=None, name=None, process_set=global_process_set):
-    """All-to-all exchange.  Returns output (and received_splits when
-    `splits` was given, matching the reference)."""
+    """Scatter slices of dim 0 to every set member and gather theirs.
+
+    splits: per-destination row counts (int sequence of set size); None =
+    uniform split (dim 0 must divide evenly).  Returns the received tensor,
+    plus received_splits when `splits` was given (reference:
+    horovod/torch/mpi_ops.py alltoall; grad = reverse exchange)."""
     output, received = HorovodAlltoall.apply(tensor, splits, name, process_set)
     if splits is None:
         return output
@@ -557,6 +581,10 @@ class HorovodReducescatter(torch.autograd.Function):
 
 def reducescatter(tensor, op=None, name=None, prescale_factor=1.0,
                   postscale_factor=1.0, process_set=global_process_set):
+    """Reduce `tensor` across the set, returning this rank's dim-0 shard
+    (rows split as evenly as possible, earlier ranks get the remainder).
+    op defaults to Average.  Reference: horovod/torch/mpi_ops.py
+    reducescatter."""
     return HorovodReducescatter.apply(tensor, op, name, prescale_factor,
                                       postscale_factor, process_set)
 
@@ -601,6 +629,8 @@ def stop_timeline():
 
 
 def barrier(process_set=global_process_set):
+    """Block until every member of the set has entered the barrier
+    (negotiation-synchronized; reference: operations.cc EnqueueBarrier)."""
     h = _core.barrier_async(_set_id(process_set))
     _register(h)
     synchronize(h)
