@@ -469,3 +469,11 @@ def test_reducescatter_integer_average_np2():
         mine = full[:2] if rank == 0 else full[2:]
         assert out.tolist() == mine, (out, mine)
     """)
+
+
+def test_bool_allreduce_np2():
+    run_workers(2, """
+        t = torch.tensor([True, False, bool(rank)])
+        out = hvd.allreduce(t, average=False, name="bool")
+        assert out.tolist() == [True, False, True], out
+    """)
