@@ -304,3 +304,19 @@ def test_integer_average_gpu(hvd):
                          average=True, name="giavg2")
     assert torch.equal(out2, torch.arange(64, device="cuda",
                                           dtype=torch.int32))
+
+
+@requires_gpu
+def test_timeline_gpu_ops(hvd, tmp_path):
+    """Timeline + roctx around real RCCL ops must not disturb results."""
+    tl = str(tmp_path / "gtl.json")
+    hvd.start_timeline(tl)
+    t = torch.randn(4096, device="cuda")
+    for i in range(5):
+        out = hvd.allreduce(t, average=False, name=f"gtl{i}")
+    hvd.stop_timeline()
+    assert torch.allclose(out, t)
+    import json
+    data = json.load(open(tl))
+    names = {str(e.get("name")) for e in data}
+    assert "ALLREDUCE" in names and "NEGOTIATE" in names, names
