@@ -32,7 +32,8 @@ bool SameSignature(const Request& a, const Request& b) {
   return a.type == b.type && a.dtype == b.dtype && a.shape == b.shape &&
          a.root_rank == b.root_rank && a.reduce_op == b.reduce_op &&
          a.process_set_id == b.process_set_id && a.device == b.device &&
-         a.group_key == b.group_key && a.group_size == b.group_size;
+         a.group_key == b.group_key && a.group_size == b.group_size &&
+         a.splits == b.splits;
 }
 
 // merge single-tensor responses of one group into one fused response
@@ -108,11 +109,20 @@ void ResponseCache::Evict(int slot) {
   free_slots_.push_back(slot);
 }
 
-void ResponseCache::Put(const Response& response, const std::vector<Request>& reqs) {
+void ResponseCache::Put(const Response& response, const std::vector<Request>& reqs,
+                        int32_t my_local_index) {
   if (response.type == ResponseType::JOIN || response.type == ResponseType::BARRIER ||
-      response.type == ResponseType::ERROR || response.type == ResponseType::TUNE ||
-      response.type == ResponseType::ALLTOALL)  // alltoall renegotiates
+      response.type == ResponseType::ERROR || response.type == ResponseType::TUNE)
     return;
+  // ALLTOALL caches only with a resolvable set-local index (the signature
+  // needs this rank's send-split row out of the full matrix)
+  int n_set = 0;
+  if (response.type == ResponseType::ALLTOALL) {
+    if (my_local_index < 0 || response.names.size() != 1) return;
+    // tensor_sizes is the n x n split matrix
+    while ((size_t)(n_set * n_set) < response.tensor_sizes.size()) ++n_set;
+    if ((size_t)(n_set * n_set) != response.tensor_sizes.size()) return;
+  }
   // Split a fused response into single-tensor cache entries.
   size_t nsizes_per = response.names.size()
                           ? response.tensor_sizes.size() / response.names.size()
@@ -152,6 +162,12 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
     sig.device = response.device;
     sig.group_key = response.group_key;
     sig.group_size = response.group_size;
+    if (response.type == ResponseType::ALLTOALL) {
+      sig.type = RequestType::ALLTOALL;
+      sig.splits.assign(
+          response.tensor_sizes.begin() + (size_t)my_local_index * n_set,
+          response.tensor_sizes.begin() + (size_t)(my_local_index + 1) * n_set);
+    }
 
     std::string key = SetKey(sig.process_set_id, sig.name);
     auto it = name_to_slot_.find(key);
@@ -209,10 +225,10 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
   std::vector<int> my_invalid_slots;
   for (auto& req : new_requests) {
     if (tl) tl->NegotiateStart(req.name);
-    if (req.type == RequestType::JOIN || req.type == RequestType::BARRIER ||
-        req.type == RequestType::ALLTOALL) {
-      // never cached: alltoall carries runtime splits; join/barrier are
-      // stateful.
+    if (req.type == RequestType::JOIN || req.type == RequestType::BARRIER) {
+      // never cached: join/barrier are stateful.  (alltoall caches too —
+      // the send-split row is part of the signature, so changed splits
+      // invalidate and renegotiate.)
       slow.push_back(std::move(req));
       continue;
     }
@@ -371,7 +387,12 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
         if (resp.type == ResponseType::JOIN || resp.type == ResponseType::BARRIER)
           inflight_.erase(SetKey(resp.process_set_id,
                                  resp.type == ResponseType::JOIN ? "join" : "barrier"));
-        cache_.Put(resp, {});
+        {
+          int32_t li = -1;
+          auto its = process_sets_.find(resp.process_set_id);
+          if (its != process_sets_.end()) li = its->second.local_index(rank_);
+          cache_.Put(resp, {}, li);
+        }
         result.responses.push_back(std::move(resp));
       }
     }

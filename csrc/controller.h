@@ -70,8 +70,10 @@ class ResponseCache {
   const Response& Get(int slot) const;
   const Request& GetRequest(int slot) const;
   // Insert/update from an executed (possibly fused) response; deterministic
-  // across ranks.  Single-tensor granularity.
-  void Put(const Response& response, const std::vector<Request>& reqs);
+  // across ranks.  Single-tensor granularity.  my_local_index: this rank's
+  // index in the response's process set (alltoall split-row extraction).
+  void Put(const Response& response, const std::vector<Request>& reqs,
+           int32_t my_local_index = -1);
   void Evict(int slot);
   size_t num_slots() const { return entries_.size(); }
   bool slot_live(int slot) const { return entries_[slot].live; }

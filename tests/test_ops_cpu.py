@@ -477,3 +477,24 @@ def test_bool_allreduce_np2():
         out = hvd.allreduce(t, average=False, name="bool")
         assert out.tolist() == [True, False, True], out
     """)
+
+
+def test_alltoall_cached_steady_state_np2():
+    """Repeated alltoall with unchanged splits takes the cache fast path;
+    changing splits invalidates and renegotiates correctly."""
+    run_workers(2, """
+        for i in range(30):
+            t = torch.arange(4).float() + rank * 10 + i
+            out, rs = hvd.alltoall(t, splits=[2, 2], name="a2ac")
+            exp = ([0.0 + i, 1 + i, 10 + i, 11 + i] if rank == 0
+                   else [2.0 + i, 3 + i, 12 + i, 13 + i])
+            assert out.tolist() == exp, (i, out, exp)
+        # changed splits -> signature invalidation -> still correct
+        splits = [1, 3] if rank == 0 else [3, 1]
+        t = torch.arange(4).float() + rank * 10
+        out, rs = hvd.alltoall(t, splits=splits, name="a2ac")
+        if rank == 0:
+            assert out.tolist() == [0.0, 10.0, 11.0, 12.0], out
+        else:
+            assert out.tolist() == [1.0, 2.0, 3.0, 13.0], out
+    """)
