@@ -101,6 +101,12 @@ int ResponseCache::AllocSlot() {
   return lru >= 0 ? lru : 0;
 }
 
+void ResponseCache::EvictSet(int32_t process_set_id) {
+  for (int i = 0; i < (int)entries_.size(); ++i)
+    if (entries_[i].live && entries_[i].request.process_set_id == process_set_id)
+      Evict(i);
+}
+
 void ResponseCache::Evict(int slot) {
   Entry& e = entries_[slot];
   if (!e.live) return;
@@ -215,7 +221,12 @@ int32_t Controller::AddProcessSet(const std::vector<int32_t>& ranks) {
   return next_set_id_ - 1;
 }
 
-void Controller::RemoveProcessSet(int32_t id) { process_sets_.erase(id); }
+void Controller::RemoveProcessSet(int32_t id) {
+  // evict the set's cache slots: stale live slots for an unknown set would
+  // win the fast-path vote forever (every rank votes non-member = ready)
+  cache_.EvictSet(id);
+  process_sets_.erase(id);
+}
 
 ResponseList Controller::RunCycle(std::vector<Request> new_requests,
                                   bool shutdown_requested) {

@@ -75,6 +75,7 @@ class ResponseCache {
   void Put(const Response& response, const std::vector<Request>& reqs,
            int32_t my_local_index = -1);
   void Evict(int slot);
+  void EvictSet(int32_t process_set_id);
   size_t num_slots() const { return entries_.size(); }
   bool slot_live(int slot) const { return entries_[slot].live; }
 

@@ -498,3 +498,20 @@ def test_alltoall_cached_steady_state_np2():
         else:
             assert out.tolist() == [1.0, 2.0, 3.0, 13.0], out
     """)
+
+
+def test_remove_process_set_evicts_cache_np2():
+    """Using a set (cached), removing it, then continuing global work must
+    not produce phantom fast-path responses (busy-spin regression guard)."""
+    run_workers(2, """
+        import time
+        ps = hvd.add_process_set([0, 1])
+        for i in range(5):
+            hvd.allreduce(torch.ones(4), average=False, process_set=ps,
+                          name="cachedsub")
+        hvd.remove_process_set(ps)
+        for i in range(10):
+            out = hvd.allreduce(torch.ones(2), average=False, name=f"g{i}")
+            assert out.sum().item() == 4.0
+        time.sleep(0.3)  # idle: bg loop must be able to sleep (no spin)
+    """)
