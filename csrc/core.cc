@@ -575,6 +575,9 @@ void PerformOperation(GlobalState& st, Response& resp) {
   try {
     if (gpu_op) {
       gpu::Execute(st, resp, entries);  // async; completion via finalizer
+      // close the submit span; the finalizer's Activity records the actual
+      // device-side completion window
+      if (tl) tl->OpEnd(resp);
     } else {
       switch (resp.type) {
         case ResponseType::ALLREDUCE:
