@@ -188,8 +188,8 @@ def wait(handle):
 def _allreduce_async_impl(tensors, outputs, name, true_op, pre, post, ps_id,
                           wire_code, kind="op", post_divisor=None):
     names = list(name) if isinstance(name, (list, tuple)) else [name]
-    h = _core.allreduce_async(tensors, outputs, names, true_op, pre, post,
-                              ps_id, wire_code)
+    h = _translate_error(_core.allreduce_async, tensors, outputs, names,
+                         true_op, pre, post, ps_id, wire_code)
     return _register(h, len(tensors), kind, post_divisor)
 
 
@@ -390,8 +390,8 @@ def sparse_allreduce_async(tensor, name, op, process_set=global_process_set):
 # ---------------------------------------------------------------------------
 def allgather_async(tensor, name=None, process_set=global_process_set):
     name = name or _next_name("allgather")
-    h = _core.allgather_async(tensor.contiguous(), "allgather." + name,
-                              _set_id(process_set))
+    h = _translate_error(_core.allgather_async, tensor.contiguous(),
+                         "allgather." + name, _set_id(process_set))
     return _register(h)
 
 
@@ -446,8 +446,8 @@ def broadcast_async(tensor, root_rank, name=None,
         tensor = tensor.contiguous()
     output = torch.empty_like(tensor)
     name = name or _next_name("broadcast")
-    h = _core.broadcast_async(tensor, output, root_rank, "broadcast." + name,
-                              _set_id(process_set))
+    h = _translate_error(_core.broadcast_async, tensor, output, root_rank,
+                         "broadcast." + name, _set_id(process_set))
     return _register(h)
 
 
@@ -456,8 +456,8 @@ def broadcast_async_(tensor, root_rank, name=None,
     if not _dense_ok(tensor):
         raise ValueError("hvd.broadcast_ requires a dense tensor")
     name = name or _next_name("broadcast")
-    h = _core.broadcast_async(tensor, tensor, root_rank, "broadcast." + name,
-                              _set_id(process_set))
+    h = _translate_error(_core.broadcast_async, tensor, tensor, root_rank,
+                         "broadcast." + name, _set_id(process_set))
     return _register(h)
 
 
@@ -506,8 +506,8 @@ def alltoall_async(tensor, splits=None, name=None,
         splits_t = torch.full((n,), first // n, dtype=torch.int64)
     else:
         splits_t = torch.as_tensor(splits, dtype=torch.int64).cpu()
-    h = _core.alltoall_async(tensor.contiguous(), splits_t, "alltoall." + name,
-                             _set_id(process_set))
+    h = _translate_error(_core.alltoall_async, tensor.contiguous(), splits_t,
+                         "alltoall." + name, _set_id(process_set))
     return _register(h, kind="alltoall_splits")
 
 
@@ -556,8 +556,9 @@ def reducescatter_async(tensor, op=None, name=None, prescale_factor=1.0,
         post_div = round(1.0 / post)  # integer Average: floor-divide post-op
         post = 1.0
     name = name or _next_name("reducescatter")
-    h = _core.reducescatter_async(tensor.contiguous(), "reducescatter." + name,
-                                  true_op, pre, post, _set_id(process_set))
+    h = _translate_error(_core.reducescatter_async, tensor.contiguous(),
+                         "reducescatter." + name, true_op, pre, post,
+                         _set_id(process_set))
     return _register(h, post_divisor=post_div)
 
 
