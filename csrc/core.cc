@@ -705,7 +705,10 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
   st.shutdown_requested = false;
   st.shutting_down = false;
   st.abort_reason.clear();
-  st.comm.Init(rank, size, addr, port);
+  double start_timeout = 120.0;
+  if (const char* e = std::getenv("HOROVOD_START_TIMEOUT"))
+    start_timeout = atof(e);
+  st.comm.Init(rank, size, addr, port, start_timeout);
   st.controller.reset(new Controller(&st.comm, rank, size, cfg));
   if (cfg.timeline_enabled) {
     const char* tf = std::getenv("HOROVOD_TIMELINE");

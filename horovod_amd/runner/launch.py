@@ -212,6 +212,8 @@ def main(argv=None):
         env["NCCL_SOCKET_IFNAME"] = args.network_interface
     if args.log_level:
         env["HOROVOD_LOG_LEVEL"] = args.log_level
+    if args.start_timeout:
+        env["HOROVOD_START_TIMEOUT"] = str(args.start_timeout)
 
     hosts = args.hosts
     if args.hostfile:
