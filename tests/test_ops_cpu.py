@@ -515,3 +515,47 @@ def test_remove_process_set_evicts_cache_np2():
             assert out.sum().item() == 4.0
         time.sleep(0.3)  # idle: bg loop must be able to sleep (no spin)
     """)
+
+
+def test_cross_check_vs_torch_distributed_np2():
+    """Independent oracle: every collective compared against
+    torch.distributed (gloo) on the same random tensors."""
+    import os as _os
+    from horovod_amd.runner.launch import find_free_port
+    port = find_free_port()
+    run_workers(2, f"""
+        import os
+        import torch.distributed as dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ["MASTER_PORT"] = "{port}"
+        dist.init_process_group("gloo", rank=rank, world_size=size)
+        torch.manual_seed(500 + rank)
+        for trial in range(5):
+            t = torch.randn(64, 3)
+            # allreduce
+            ours = hvd.allreduce(t, average=False, name=f"xc.ar.{{trial}}")
+            ref = t.clone(); dist.all_reduce(ref)
+            assert torch.allclose(ours, ref, atol=1e-6), "allreduce"
+            # allgather
+            ours = hvd.allgather(t, name=f"xc.ag.{{trial}}")
+            outs = [torch.empty_like(t) for _ in range(size)]
+            dist.all_gather(outs, t)
+            assert torch.allclose(ours, torch.cat(outs), atol=1e-6), "ag"
+            # broadcast
+            b = t.clone(); ours = hvd.broadcast(b, root_rank=1,
+                                                name=f"xc.bc.{{trial}}")
+            ref = t.clone(); dist.broadcast(ref, src=1)
+            assert torch.allclose(ours, ref, atol=1e-6), "bcast"
+            # reducescatter (sum)
+            ours = hvd.reducescatter(t, op=hvd.Sum, name=f"xc.rs.{{trial}}")
+            full = t.clone(); dist.all_reduce(full)
+            assert torch.allclose(ours, full[rank*32:(rank+1)*32],
+                                  atol=1e-6), "rs"
+            # alltoall
+            ours, _ = hvd.alltoall(t, splits=[32, 32],
+                                   name=f"xc.a2a.{{trial}}")
+            outs = [torch.empty(32, 3) for _ in range(size)]
+            dist.all_to_all(outs, list(t.chunk(2)))
+            assert torch.allclose(ours, torch.cat(outs), atol=1e-6), "a2a"
+        dist.destroy_process_group()
+    """, timeout=300)
