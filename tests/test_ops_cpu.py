@@ -551,11 +551,12 @@ def test_cross_check_vs_torch_distributed_np2():
             full = t.clone(); dist.all_reduce(full)
             assert torch.allclose(ours, full[rank*32:(rank+1)*32],
                                   atol=1e-6), "rs"
-            # alltoall
+            # alltoall (gloo lacks all_to_all: oracle via all_gather)
             ours, _ = hvd.alltoall(t, splits=[32, 32],
                                    name=f"xc.a2a.{{trial}}")
-            outs = [torch.empty(32, 3) for _ in range(size)]
-            dist.all_to_all(outs, list(t.chunk(2)))
-            assert torch.allclose(ours, torch.cat(outs), atol=1e-6), "a2a"
+            everyone = [torch.empty_like(t) for _ in range(size)]
+            dist.all_gather(everyone, t)
+            expected = torch.cat([e.chunk(2)[rank] for e in everyone])
+            assert torch.allclose(ours, expected, atol=1e-6), "a2a"
         dist.destroy_process_group()
     """, timeout=300)
