@@ -560,3 +560,12 @@ def test_cross_check_vs_torch_distributed_np2():
             assert torch.allclose(ours, expected, atol=1e-6), "a2a"
         dist.destroy_process_group()
     """, timeout=300)
+
+
+def test_alltoall_int64_np2():
+    run_workers(2, """
+        t = torch.arange(4, dtype=torch.int64) + rank * 100
+        out, rs = hvd.alltoall(t, splits=[2, 2], name="a2ai")
+        exp = [0, 1, 100, 101] if rank == 0 else [2, 3, 102, 103]
+        assert out.tolist() == exp, out
+    """)

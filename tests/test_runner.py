@@ -107,3 +107,17 @@ def test_check_build(capsys):
     assert main(["--check-build"]) == 0
     out = capsys.readouterr().out
     assert "PyTorch (ROCm)" in out and "RCCL" in out
+
+
+def test_interactive_run_propagates_failure():
+    import horovod_amd
+
+    def bad():
+        import horovod_amd.torch as hvd
+        hvd.init()
+        if hvd.rank() == 1:
+            raise RuntimeError("worker boom")
+        return 1
+
+    with pytest.raises(RuntimeError):
+        horovod_amd.run(bad, np=2)
