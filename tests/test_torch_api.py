@@ -520,3 +520,32 @@ def test_optimizer_groups_with_compression_np2():
         assert torch.allclose(g[0], g[1], atol=1e-5), \
             (g[0] - g[1]).abs().max()
     """)
+
+
+def test_adasum_golden_np5():
+    rng = np.random.RandomState(3)
+    vecs = [rng.randn(16).astype(np.float32) for _ in range(5)]
+    expected = _adasum_golden(vecs)
+    run_workers(5, f"""
+        import numpy as np
+        allv = {[v.tolist() for v in vecs]!r}
+        t = torch.tensor(allv[rank], dtype=torch.float32)
+        out = hvd.allreduce(t, op=hvd.Adasum, name="adasum5")
+        expected = np.array({expected.tolist()!r})
+        assert np.allclose(out.numpy(), expected, atol=1e-5)
+    """)
+
+
+def test_collectives_np5():
+    run_workers(5, """
+        out = hvd.allreduce(torch.ones(3) * (rank + 1), average=False,
+                            name="n5")
+        assert out[0].item() == 15.0
+        g = hvd.allgather(torch.full((1,), float(rank)), name="n5g")
+        assert g.tolist() == [0.0, 1.0, 2.0, 3.0, 4.0]
+        rs = hvd.reducescatter(torch.arange(7).float(), op=hvd.Sum,
+                               name="n5rs")
+        # 7 rows over 5 ranks: [2,2,1,1,1]
+        expected_rows = [2, 2, 1, 1, 1][rank]
+        assert rs.numel() == expected_rows, rs
+    """)
