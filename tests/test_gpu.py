@@ -320,3 +320,15 @@ def test_timeline_gpu_ops(hvd, tmp_path):
     data = json.load(open(tl))
     names = {str(e.get("name")) for e in data}
     assert "ALLREDUCE" in names and "NEGOTIATE" in names, names
+
+
+@requires_gpu
+def test_mixed_cpu_gpu_ops(hvd):
+    """Interleaved CPU-tensor and GPU-tensor collectives in one session
+    (separate fusion classes; both paths live)."""
+    for i in range(5):
+        c = hvd.allreduce(torch.ones(64) * i, average=False, name=f"mcpu{i}")
+        g = hvd.allreduce(torch.ones(64, device="cuda") * i, average=False,
+                          name=f"mgpu{i}")
+        assert c.sum().item() == 64.0 * i
+        assert g.sum().item() == 64.0 * i
