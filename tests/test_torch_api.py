@@ -549,3 +549,26 @@ def test_collectives_np5():
         expected_rows = [2, 2, 1, 1, 1][rank]
         assert rs.numel() == expected_rows, rs
     """)
+
+
+def test_groups_with_backward_passes_np2():
+    run_workers(2, """
+        torch.manual_seed(17)
+        model = torch.nn.Sequential(torch.nn.Linear(4, 6), torch.nn.ReLU(),
+                                    torch.nn.Linear(6, 2))
+        opt = hvd.DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1),
+            named_parameters=model.named_parameters(), groups=2,
+            backward_passes_per_step=2)
+        hvd.broadcast_parameters(model.state_dict(), root_rank=0)
+        torch.manual_seed(60 + rank)
+        for step in range(2):
+            opt.zero_grad()
+            for micro in range(2):
+                model(torch.randn(3, 4)).sum().backward()
+            opt.step()
+        flat = torch.cat([p.detach().flatten() for p in model.parameters()])
+        g = hvd.allgather(flat.unsqueeze(0), name="gbp")
+        assert torch.allclose(g[0], g[1], atol=1e-6), \
+            (g[0] - g[1]).abs().max()
+    """)
