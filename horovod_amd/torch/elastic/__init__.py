@@ -1,7 +1,8 @@
 from horovod_amd.torch.elastic.state import TorchState  # noqa: F401
 from horovod_amd.torch.elastic.sampler import ElasticSampler  # noqa: F401
 
-from horovod_amd.common.elastic import run_fn, _rendezvous_reset
+from horovod_amd.common.elastic import (ObjectState, State,  # noqa: F401
+                                        run_fn, _rendezvous_reset)
 
 
 def run(func):
