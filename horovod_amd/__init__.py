@@ -7,6 +7,13 @@ Structural reference: horovod/horovod v0.28.1.  This is a
 from-scratch design, not a port: single GPU data plane (RCCL), TCP star
 control plane (no MPI/Gloo), torch-only front-end.
 """
+# torch must load before the native extension: _core.so links libtorch, and
+# letting the dynamic loader pull libtorch in WITHOUT the python module's
+# own initialization corrupts interpreter teardown (glibc fastbin abort at
+# exit, observed under pytest).  Importing torch here guarantees ordering
+# for any `import horovod_amd._core`.
+import torch  # noqa: F401  (ordering dependency, see above)
+
 __version__ = "0.1.0"
 
 
