@@ -344,12 +344,17 @@ def _grouped_allreduce_impl(tensors, outputs, average, name, op,
                                  post_divisor=post_div)
 
 
-def grouped_allreduce(tensors, average=None, name=None, op=None,
-                      prescale_factor=1.0, postscale_factor=1.0,
+def grouped_allreduce(tensors, average=None, name=None, compression=None,
+                      op=None, prescale_factor=1.0, postscale_factor=1.0,
                       process_set=global_process_set):
-    handle = grouped_allreduce_async(tensors, average, name, op,
+    from horovod_amd.torch.compression import Compression
+    wire = None
+    if compression is not None and compression is not Compression.none:
+        wire = compression.wire_dtype(tensors[0].dtype)
+    outputs = [torch.empty_like(t) for t in tensors]
+    handle = _grouped_allreduce_impl(tensors, outputs, average, name, op,
                                      prescale_factor, postscale_factor,
-                                     process_set)
+                                     process_set, wire_dtype=wire)
     return synchronize(handle)
 
 

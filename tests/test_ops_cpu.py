@@ -569,3 +569,27 @@ def test_alltoall_int64_np2():
         exp = [0, 1, 100, 101] if rank == 0 else [2, 3, 102, 103]
         assert out.tolist() == exp, out
     """)
+
+
+def test_concurrent_sets_grouped_compressed_np4():
+    """Two overlapping process sets with grouped + wire-compressed ops
+    interleaved with global collectives."""
+    run_workers(4, """
+        from horovod_amd.torch.compression import Compression
+        even = hvd.add_process_set([0, 2])
+        odd = hvd.add_process_set([1, 3])
+        mine = even if rank % 2 == 0 else odd
+        peer_sum = (rank % 2 + 1) + (rank % 2 + 3)  # ranks+1 summed in my set
+        for i in range(10):
+            ts = [torch.ones(100) * (rank + 1), torch.ones(50) * (rank + 1)]
+            from horovod_amd.torch.compression import Compression as _C
+            comp = _C.fp16 if i % 2 else None
+            outs = hvd.grouped_allreduce(ts, average=False, process_set=mine,
+                                         name=f"cs{i}{bool(comp)}",
+                                         compression=comp)
+            assert outs[0][0].item() == peer_sum, (i, outs[0][0])
+            g = hvd.allreduce(torch.ones(10) * (rank + 1), average=False,
+                              name=f"cg{i}", compression=Compression.fp16)
+            assert abs(g[0].item() - 10.0) < 0.1, g[0]
+        hvd.barrier()
+    """)
