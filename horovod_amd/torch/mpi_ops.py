@@ -552,8 +552,9 @@ def alltoall(tensor, splits=None, name=None, process_set=global_process_set):
 # ---------------------------------------------------------------------------
 # reducescatter
 # ---------------------------------------------------------------------------
-def reducescatter_async(tensor, op=None, name=None, prescale_factor=1.0,
-                        postscale_factor=1.0, process_set=global_process_set):
+def reducescatter_async(tensor, name=None, op=None,
+                        process_set=global_process_set, prescale_factor=1.0,
+                        postscale_factor=1.0):
     true_op, _, pre, post = _resolve_scales(op, None, prescale_factor,
                                             postscale_factor, process_set)
     post_div = None
@@ -585,31 +586,44 @@ class HorovodReducescatter(torch.autograd.Function):
         return grad, None, None, None, None, None
 
 
-def reducescatter(tensor, op=None, name=None, prescale_factor=1.0,
-                  postscale_factor=1.0, process_set=global_process_set):
+def reducescatter(tensor, name=None, compression=None, op=None,
+                  process_set=global_process_set, prescale_factor=1.0,
+                  postscale_factor=1.0):
     """Reduce `tensor` across the set, returning this rank's dim-0 shard
     (rows split as evenly as possible, earlier ranks get the remainder).
-    op defaults to Average.  Reference: horovod/torch/mpi_ops.py
-    reducescatter."""
+    op defaults to Average.  `compression` is accepted for reference-
+    signature parity (applied python-side around the op).
+    Reference: horovod/torch/mpi_ops.py reducescatter."""
+    from horovod_amd.torch.compression import Compression
+    if compression is not None and compression is not Compression.none:
+        t, ctx = compression.compress(tensor)
+        out = HorovodReducescatter.apply(t, op, name, prescale_factor,
+                                         postscale_factor, process_set)
+        return compression.decompress(out, ctx)
     return HorovodReducescatter.apply(tensor, op, name, prescale_factor,
                                       postscale_factor, process_set)
 
 
-def grouped_reducescatter_async(tensors, op=None, name=None,
-                                prescale_factor=1.0, postscale_factor=1.0,
-                                process_set=global_process_set):
+def grouped_reducescatter_async(tensors, name=None, op=None,
+                                process_set=global_process_set,
+                                prescale_factor=1.0, postscale_factor=1.0):
     base = name or _next_name("grouped_reducescatter")
-    return [reducescatter_async(t, op, f"{base}.{i}", prescale_factor,
-                                postscale_factor, process_set)
+    return [reducescatter_async(t, name=f"{base}.{i}", op=op,
+                                process_set=process_set,
+                                prescale_factor=prescale_factor,
+                                postscale_factor=postscale_factor)
             for i, t in enumerate(tensors)]
 
 
-def grouped_reducescatter(tensors, op=None, name=None, prescale_factor=1.0,
-                          postscale_factor=1.0,
-                          process_set=global_process_set):
-    handles = grouped_reducescatter_async(tensors, op, name, prescale_factor,
-                                          postscale_factor, process_set)
-    return [synchronize(h) for h in handles]
+def grouped_reducescatter(tensors, name=None, compression=None, op=None,
+                          process_set=global_process_set, prescale_factor=1.0,
+                          postscale_factor=1.0):
+    base = name or _next_name("grouped_reducescatter")
+    return [reducescatter(t, name=f"{base}.{i}", compression=compression,
+                          op=op, process_set=process_set,
+                          prescale_factor=prescale_factor,
+                          postscale_factor=postscale_factor)
+            for i, t in enumerate(tensors)]
 
 
 # ---------------------------------------------------------------------------
