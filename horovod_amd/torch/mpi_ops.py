@@ -574,8 +574,10 @@ class HorovodReducescatter(torch.autograd.Function):
                 process_set):
         ctx.process_set = process_set
         ctx.op = op
-        handle = reducescatter_async(tensor, op, name, prescale_factor,
-                                     postscale_factor, process_set)
+        handle = reducescatter_async(tensor, name=name, op=op,
+                                     process_set=process_set,
+                                     prescale_factor=prescale_factor,
+                                     postscale_factor=postscale_factor)
         return synchronize(handle)
 
     @staticmethod
