@@ -69,9 +69,9 @@ class ElasticSampler(torch.utils.data.Sampler):
             "processed_indices": sorted(self.processed_indices),
         }
 
-    def load_state_dict(self, state):
-        self.epoch = state["epoch"]
-        self.processed_indices = set(state["processed_indices"])
+    def load_state_dict(self, state_dict):
+        self.epoch = state_dict["epoch"]
+        self.processed_indices = set(state_dict["processed_indices"])
 
     def __iter__(self):
         return iter(self.indices)

@@ -76,12 +76,14 @@ def allgather_object(obj, name=None, process_set=global_process_set):
     return out
 
 
-def broadcast_optimizer_state(optimizer, root_rank,
+def broadcast_optimizer_state(optimizer, root_rank, model=None,
                               process_set=global_process_set):
     """Broadcast an optimizer's state from root (reference:
     functions.py:74-199).  Tensor state entries are broadcast in place;
     non-tensor entries (step counters, hyperparameters) travel via
     broadcast_object."""
+    del model  # accepted for reference-signature parity (state is derived
+    # from the optimizer's param_groups directly here)
     if isinstance(optimizer, torch.optim.LBFGS):
         raise ValueError("cannot broadcast torch.optim.LBFGS state")
 

@@ -205,6 +205,19 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._synchronized = False
         return super(self.__class__, self).step(closure)
 
+    def set_backward_passes_per_step(self, passes):
+        """Reference: optimizer.py set_backward_passes_per_step."""
+        self.backward_passes_per_step = passes
+        for p in self._allreduce_delay:
+            self._allreduce_delay[p] = passes
+
+    def load_state_dict(self, *args, **kwargs):
+        # dropping in a new state invalidates in-flight gradient bookkeeping
+        # (reference: _DistributedOptimizer.load_state_dict)
+        result = super(self.__class__, self).load_state_dict(*args, **kwargs)
+        self.reset_distributed_state()
+        return result
+
     def reset_distributed_state(self):
         """Drop in-flight allreduce bookkeeping after an elastic reset (the
         native handles died with the old controller)."""
@@ -282,6 +295,19 @@ class _DistributedAdasumOptimizer(torch.optim.Optimizer):
 
     def zero_grad(self, *args, **kwargs):
         return super(self.__class__, self).zero_grad(*args, **kwargs)
+
+    def set_backward_passes_per_step(self, passes):
+        """Reference: optimizer.py set_backward_passes_per_step."""
+        self.backward_passes_per_step = passes
+        for p in self._allreduce_delay:
+            self._allreduce_delay[p] = passes
+
+    def load_state_dict(self, *args, **kwargs):
+        # dropping in a new state invalidates in-flight gradient bookkeeping
+        # (reference: _DistributedOptimizer.load_state_dict)
+        result = super(self.__class__, self).load_state_dict(*args, **kwargs)
+        self.reset_distributed_state()
+        return result
 
     def reset_distributed_state(self):
         pass
