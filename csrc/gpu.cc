@@ -434,8 +434,8 @@ void ExecuteAdasum(GlobalState& st, DeviceCtx& ctx, Response& resp,
   }
   for (auto& e : entries)
     if (!e.output.defined()) {
-      c10::hip::HIPStreamGuard sg(ctx.stream);
       e.output = at::empty_like(e.tensor);
+      RecordStreamFor(e.output, ctx.stream);
     }
   PackEntries(ctx, entries, wire, true, result_base);
 }
@@ -529,8 +529,10 @@ void Execute(GlobalState& st, Response& resp,
                                  ToNcclOp(resp.reduce_op), comm, stream));
         for (auto& e : entries)
           if (!e.output.defined()) {
-            c10::hip::HIPStreamGuard sg(ctx.stream);
+            // allocate on the DEFAULT stream (the consumer side) and record
+            // the comm-stream use: frees then correctly fence both streams
             e.output = at::empty_like(e.tensor);
+            RecordStreamFor(e.output, ctx.stream);
           }
         PackEntries(ctx, entries, wire, true);
       }
@@ -570,10 +572,8 @@ void Execute(GlobalState& st, Response& resp,
                                      e.tensor.sizes().end());
       if (out_shape.empty()) out_shape = {total0};
       else out_shape[0] = total0;
-      {
-        c10::hip::HIPStreamGuard sg(ctx.stream);
-        e.output = at::empty(out_shape, e.tensor.options());
-      }
+      e.output = at::empty(out_shape, e.tensor.options());
+      RecordStreamFor(e.output, ctx.stream);
       if (same && total0 > 0) {
         RCCL_CHECK(ncclAllGather(in.data_ptr(), e.output.data_ptr(),
                                  resp.tensor_sizes[0] * row_elems, wire_nccl, comm,
@@ -610,10 +610,8 @@ void Execute(GlobalState& st, Response& resp,
                                      e.tensor.sizes().end());
       if (out_shape.empty()) out_shape = {recv_rows};
       else out_shape[0] = recv_rows;
-      {
-        c10::hip::HIPStreamGuard sg(ctx.stream);
-        e.output = at::empty(out_shape, e.tensor.options());
-      }
+      e.output = at::empty(out_shape, e.tensor.options());
+      RecordStreamFor(e.output, ctx.stream);
       char* in_base = (char*)in.data_ptr();
       char* out_base = (char*)e.output.data_ptr();
       RCCL_CHECK(ncclGroupStart());
@@ -648,10 +646,8 @@ void Execute(GlobalState& st, Response& resp,
                                      e.tensor.sizes().end());
       if (out_shape.empty()) out_shape = {my_rows};
       else out_shape[0] = my_rows;
-      {
-        c10::hip::HIPStreamGuard sg(ctx.stream);
-        e.output = at::empty(out_shape, e.tensor.options());
-      }
+      e.output = at::empty(out_shape, e.tensor.options());
+      RecordStreamFor(e.output, ctx.stream);
       if (rem == 0 && first > 0) {
         RCCL_CHECK(ncclReduceScatter(in.data_ptr(), e.output.data_ptr(),
                                      base_rows * row_elems, wire_nccl,
