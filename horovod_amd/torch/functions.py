@@ -3,9 +3,7 @@
 Reference: horovod/torch/functions.py:30-279 — broadcast_parameters,
 broadcast_optimizer_state, broadcast_object, allgather_object.
 """
-import collections
 import io
-import pickle
 
 import cloudpickle
 import torch
@@ -71,7 +69,8 @@ def allgather_object(obj, name=None, process_set=global_process_set):
     gathered = allgather(payload, name=f"{name}.data", process_set=process_set)
     out, off = [], 0
     for s in sizes.tolist():
-        out.append(pickle.loads(gathered[off:off + s].numpy().tobytes()))
+        out.append(cloudpickle.loads(
+            gathered[off:off + s].numpy().tobytes()))
         off += s
     return out
 
