@@ -377,7 +377,8 @@ def test_scalar_and_edge_shapes_np2():
 def test_join_last_rank_np3():
     run_workers(3, """
         import time
-        time.sleep(0.2 * rank)  # rank 2 joins last
+        hvd.barrier()            # align ranks before the staggered delays
+        time.sleep(0.7 * rank)   # rank 2 joins well after the others
         last = hvd.join()
         assert last == 2, last
     """)
