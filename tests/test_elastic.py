@@ -97,7 +97,7 @@ def test_elastic_worker_failure_recovery(tmp_path):
 def test_elastic_scale_up(tmp_path):
     driver, disc, marker = _driver(
         tmp_path, {"127.0.0.1": 1},
-        extra_env={"TEST_TARGET_BATCHES": "60"})
+        extra_env={"TEST_TARGET_BATCHES": "120"})
     driver.start()
     time.sleep(2.0)
     disc.set({"127.0.0.1": 2})
