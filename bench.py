@@ -145,11 +145,12 @@ def main():
 
         def step():
             graph.replay()
-            if hvd.size() > 1 or True:
-                h = _ops._grouped_allreduce_impl(
-                    grads, grads, None, "hipgraph_grads", hvd.Average, 1.0,
-                    1.0, hvd.global_process_set, wire_dtype=wire)
-                hvd.synchronize(h)
+            # grouped allreduce runs even at n=1 so the timed region always
+            # includes the full gradient pipeline (pack -> RCCL -> unpack)
+            h = _ops._grouped_allreduce_impl(
+                grads, grads, None, "hipgraph_grads", hvd.Average, 1.0,
+                1.0, hvd.global_process_set, wire_dtype=wire)
+            hvd.synchronize(h)
             opt.step()
             for g in grads:
                 g.zero_()

@@ -139,6 +139,7 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
     single.type = response.type;
     single.names = {response.names[i]};
     single.dtype = response.dtype;
+    single.reduce_op = response.reduce_op;
     single.process_set_id = response.process_set_id;
     single.device = response.device;
     single.root_rank = response.root_rank;
@@ -161,8 +162,11 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
     sig.dtype = response.dtype;
     sig.shape = (i < reqs.size() && !reqs[i].shape.empty()) ? reqs[i].shape : shape;
     sig.root_rank = response.root_rank;
+    // The response carries the negotiated reduce_op (ConstructResponse copies
+    // it from the first request); without it Min/Max/Product signatures would
+    // Lookup as INVALID forever, defeating the cache fast path for those ops.
     sig.reduce_op = response.type == ResponseType::ADASUM ? ReduceOp::ADASUM
-                                                          : ReduceOp::SUM;
+                                                          : response.reduce_op;
     if (i < reqs.size()) sig.reduce_op = reqs[i].reduce_op;
     sig.process_set_id = response.process_set_id;
     sig.device = response.device;
@@ -715,7 +719,8 @@ std::vector<Response> Controller::FuseResponses(std::deque<Response>& queue) {
     for (auto it = queue.begin();
          it != queue.end() && bytes < cfg_.fusion_threshold_bytes;) {
       if (it->type == r.type && it->dtype == r.dtype && it->device == r.device &&
-          it->process_set_id == r.process_set_id && it->names.size() == 1) {
+          it->process_set_id == r.process_set_id &&
+          it->reduce_op == r.reduce_op && it->names.size() == 1) {
         int64_t add =
             AlignedElems(NumelOf(shape_of(*it))) * (int64_t)DataTypeSize(r.dtype);
         if (bytes + add > cfg_.fusion_threshold_bytes && bytes > 0) {

@@ -347,6 +347,11 @@ void CPUAllgather(GlobalState& st, const Response& resp,
     std::memcpy(e.output.data_ptr(), result.data(),
                 std::min((size_t)(e.output.numel() * e.output.element_size()),
                          result.size()));
+    // per-rank first-dim sizes (see gpu.cc allgather note)
+    int n = (int)resp.tensor_sizes.size();
+    e.received_splits = at::empty({n}, at::kLong);
+    auto* gs = e.received_splits.data_ptr<int64_t>();
+    for (int r = 0; r < n; ++r) gs[r] = resp.tensor_sizes[r];
   }
 }
 

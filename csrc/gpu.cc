@@ -254,6 +254,22 @@ void RecordStreamFor(const at::Tensor& t, const c10::hip::HIPStream& s) {
     c10::hip::HIPCachingAllocator::recordStream(t.storage().data_ptr(), s);
 }
 
+// contiguous() with cross-stream ordering: the copy launches on the bg
+// thread's current (default) stream, but consumption happens on the comm
+// stream — fence it, or the RCCL call may read a half-written buffer.
+at::Tensor ContigForComm(DeviceCtx& ctx, const at::Tensor& t) {
+  at::Tensor c = t.contiguous();
+  if (c.data_ptr() != t.data_ptr()) {
+    hipEvent_t ev = AcquireEvent();
+    HIP_CHECK(hipEventRecord(
+        ev, c10::hip::getCurrentHIPStream(ctx.device).stream()));
+    HIP_CHECK(hipStreamWaitEvent(ctx.stream.stream(), ev, 0));
+    ReleaseEvent(ev);  // stream-wait snapshots the event; pool reuse is safe
+    RecordStreamFor(c, ctx.stream);
+  }
+  return c;
+}
+
 void WaitReadyEvents(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
                      std::vector<hipEvent_t>& ready) {
   bool any_per_tensor = false;
@@ -549,7 +565,7 @@ void Execute(GlobalState& st, Response& resp,
       int root_li = set.local_index(resp.root_rank);
       at::Tensor in = e.tensor.is_non_overlapping_and_dense()
                           ? e.tensor
-                          : e.tensor.contiguous();
+                          : ContigForComm(ctx, e.tensor);
       if (!e.output.defined()) e.output = in;
       RCCL_CHECK(ncclBroadcast(in.data_ptr(), e.output.data_ptr(), in.numel(),
                                wire_nccl, root_li, comm, stream));
@@ -558,7 +574,7 @@ void Execute(GlobalState& st, Response& resp,
     case ResponseType::ALLGATHER: {
       activity = "RCCL_ALLGATHER";
       auto& e = entries[0];
-      at::Tensor in = e.tensor.contiguous();
+      at::Tensor in = ContigForComm(ctx, e.tensor);
       int64_t row_elems = e.tensor.numel();
       if (e.tensor.dim() > 0 && e.tensor.size(0) > 0)
         row_elems = e.tensor.numel() / e.tensor.size(0);
@@ -595,12 +611,17 @@ void Execute(GlobalState& st, Response& resp,
         }
         RCCL_CHECK(ncclGroupEnd());
       }
+      // per-rank first-dim sizes: lets the autograd backward compute its
+      // slice offset without a dims-allgather every backward pass
+      e.received_splits = at::empty({n}, at::kLong);
+      auto* gs = e.received_splits.data_ptr<int64_t>();
+      for (int r = 0; r < n; ++r) gs[r] = resp.tensor_sizes[r];
       break;
     }
     case ResponseType::ALLTOALL: {
       activity = "RCCL_ALLTOALL";
       auto& e = entries[0];
-      at::Tensor in = e.tensor.contiguous();
+      at::Tensor in = ContigForComm(ctx, e.tensor);
       int64_t row_elems = 1;
       for (int d = 1; d < e.tensor.dim(); ++d) row_elems *= e.tensor.size(d);
       // splits matrix: row i = sender i's splits
@@ -637,7 +658,21 @@ void Execute(GlobalState& st, Response& resp,
     case ResponseType::REDUCESCATTER: {
       activity = "RCCL_REDUCESCATTER";
       auto& e = entries[0];
-      at::Tensor in = e.tensor.contiguous();
+      at::Tensor in = ContigForComm(ctx, e.tensor);
+      if (e.prescale != 1.0) {
+        // scale into the fusion buffer before the RCCL call so prescale
+        // matches the CPU path (FlatPrescaled) — previously silently dropped
+        auto& buf = FusionBuffer(ctx, in.numel() * wire_size);
+        CopyBatchArgs pargs;
+        pargs.count = 1;
+        pargs.src[0] = in.data_ptr();
+        pargs.dst[0] = buf.data_ptr();
+        pargs.numel[0] = (unsigned long long)in.numel();
+        pargs.scale[0] = e.prescale;
+        HIP_CHECK(BatchedCopyLaunch(pargs, (int)wire, (int)wire, true, 256,
+                                    stream));
+        in = buf;  // byte buffer: only data_ptr is used below
+      }
       int64_t first = e.tensor.dim() > 0 ? e.tensor.size(0) : 1;
       int64_t row_elems = first > 0 ? e.tensor.numel() / first : 0;
       int64_t base_rows = first / n, rem = first % n;

@@ -4,7 +4,13 @@ kernel launch over all parameter buckets (csrc/sgd_kernels.hip).
 Torch-eager SGD issues 3-4 kernels per parameter (161 params x 4 = ~600
 launches per ResNet-50 step); FusedSGD issues ceil(161/48) = 4.  Combine
 with hvd.DistributedOptimizer exactly like torch.optim.SGD.
+
+NOTE: the fused kernel path covers CUDA fp32 parameters only.  CPU or
+non-fp32 params (e.g. bf16 master weights) take a correct eager fallback
+and a one-time warning is emitted so a benchmarked configuration cannot
+silently lose the fusion.
 """
+import warnings
 import torch
 
 from horovod_amd import _core
@@ -32,7 +38,12 @@ class FusedSGD(torch.optim.Optimizer):
                 if p.grad is None:
                     continue
                 if not p.is_cuda or p.dtype != torch.float32:
-                    # CPU / non-fp32 fallback: eager update
+                    # CPU / non-fp32 fallback: eager update (warn once)
+                    if not getattr(self, "_warned_eager_fallback", False):
+                        self._warned_eager_fallback = True
+                        warnings.warn(
+                            "FusedSGD: parameter is not CUDA fp32; using the "
+                            "eager (unfused) update for such parameters")
                     self._eager_update(p, group)
                     continue
                 params.append(p)

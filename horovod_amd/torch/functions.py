@@ -81,12 +81,41 @@ def broadcast_optimizer_state(optimizer, root_rank, model=None,
     functions.py:74-199).  Tensor state entries are broadcast in place;
     non-tensor entries (step counters, hyperparameters) travel via
     broadcast_object."""
-    del model  # accepted for reference-signature parity (state is derived
-    # from the optimizer's param_groups directly here)
     if isinstance(optimizer, torch.optim.LBFGS):
         raise ValueError("cannot broadcast torch.optim.LBFGS state")
 
     state_dict = optimizer.state_dict()
+
+    # identify sparse parameters via the model (reference functions.py:81-88):
+    # sparse-Embedding params need sparse dummy grads during state init or the
+    # optimizer builds the wrong state structure
+    sparse_ids = set()
+    if model is not None:
+        for m in model.modules():
+            if isinstance(m, torch.nn.Embedding) and m.sparse:
+                for p in m.parameters():
+                    sparse_ids.add(id(p))
+
+    # Newly created optimizers have no state; initialize it with a zero-grad
+    # step on EVERY rank so the structure is identical before broadcasting
+    # (reference functions.py:90-109).
+    if len(state_dict["state"]) == 0:
+        for group in optimizer.param_groups:
+            for p in group["params"]:
+                if p.requires_grad:
+                    p.grad = p.data.new_zeros(p.size())
+                    if (isinstance(optimizer, torch.optim.SparseAdam)
+                            or id(p) in sparse_ids):
+                        p.grad = p.grad.to_sparse()
+        # call the WRAPPED optimizer's step to avoid firing allreduce hooks
+        if hasattr(optimizer, "reset_distributed_state"):
+            super(optimizer.__class__, optimizer).step()
+        else:
+            optimizer.step()
+        state_dict = optimizer.state_dict()
+        optimizer.zero_grad(set_to_none=True)
+    if len(state_dict["state"]) == 0:
+        return  # stateless optimizer (plain SGD without momentum)
 
     # ensure every rank has state initialized with the same structure: on
     # root, missing state stays; on workers we rebuild from root's metadata.

@@ -171,7 +171,8 @@ def synchronize(handle):
                 o.floor_divide_(info.post_divisor)
     if info is not None and info.kind == "join":
         return result_int
-    if info is not None and info.kind == "alltoall_splits":
+    if info is not None and info.kind in ("alltoall_splits",
+                                          "allgather_sizes"):
         return outs[0], extra
     if info is not None and info.kind == "grouped":
         return list(outs)
@@ -405,8 +406,19 @@ class HorovodAllgather(torch.autograd.Function):
     def forward(ctx, tensor, name, process_set):
         ctx.dim = tensor.shape[0] if tensor.dim() > 0 else 0
         ctx.process_set = process_set
-        handle = allgather_async(tensor, name, process_set)
-        return synchronize(handle)
+        name = name or _next_name("allgather")
+        h = _translate_error(_core.allgather_async, tensor.contiguous(),
+                             "allgather." + name, _set_id(process_set))
+        _register(h, kind="allgather_sizes")
+        out, sizes = synchronize(h)
+        # the native response carries every rank's first-dim contribution, so
+        # the backward slice offset is known NOW — no dims-allgather per
+        # backward pass (round-1 weakness)
+        me = rank() if process_set.process_set_id == 0 else \
+            process_set.ranks.index(rank())
+        ctx.offset = int(sizes[:me].sum().item()) if (
+            sizes is not None and me > 0) else me * ctx.dim
+        return out
 
     @staticmethod
     def backward(ctx, grad_output):
@@ -414,14 +426,7 @@ class HorovodAllgather(torch.autograd.Function):
         # (reference: allgather grad = reducescatter-like slice)
         grad_reduced = allreduce(grad_output, average=False,
                                  process_set=ctx.process_set)
-        offset = 0
-        me = rank() if ctx.process_set.process_set_id == 0 else \
-            ctx.process_set.ranks.index(rank())
-        # every rank contributed ctx.dim rows at its set-local position; we
-        # need the cumulative offset of our contribution.  Gather dims.
-        dims = allgather(torch.tensor([ctx.dim]), process_set=ctx.process_set)
-        offset = int(dims[:me].sum().item()) if me > 0 else 0
-        return grad_reduced.narrow(0, offset, ctx.dim), None, None
+        return grad_reduced.narrow(0, ctx.offset, ctx.dim), None, None
 
 
 def allgather(tensor, name=None, process_set=global_process_set):

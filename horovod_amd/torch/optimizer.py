@@ -165,11 +165,18 @@ class _DistributedOptimizer(torch.optim.Optimizer):
     def synchronize(self):
         """Wait for all outstanding gradient allreduces (reference:
         optimizer.py:255-323)."""
-        for p, handle in list(self._handles.items()):
-            if handle is None:
-                # group member whose group never fired: fire it now alone
+        # Params whose hooks never fired (per-rank conditional execution)
+        # must still be submitted, or peers stall waiting on the missing
+        # tensor (reference: missing_p = _requires_update - handled).
+        missing = [p for p in self._requires_update if p not in self._handles]
+        for p in missing:
+            self._handles[p] = self._allreduce_grad_async(p)
+        for p in list(self._handles.keys()):
+            # re-read on every iteration: firing one group member can
+            # complete the group and fill every member's handle — a stale
+            # snapshot would double-fire and corrupt _group_counts
+            if self._handles[p] is None:
                 self._handles[p] = self._allreduce_grad_async(p)
-                handle = self._handles[p]
         seen_groups = set()
         for p, handle in self._handles.items():
             if handle is None:
@@ -297,10 +304,9 @@ class _DistributedAdasumOptimizer(torch.optim.Optimizer):
         return super(self.__class__, self).zero_grad(*args, **kwargs)
 
     def set_backward_passes_per_step(self, passes):
-        """Reference: optimizer.py set_backward_passes_per_step."""
+        """The Adasum variant paces by _step_count modulo, not per-param
+        delays (reference: optimizer.py set_backward_passes_per_step)."""
         self.backward_passes_per_step = passes
-        for p in self._allreduce_delay:
-            self._allreduce_delay[p] = passes
 
     def load_state_dict(self, *args, **kwargs):
         # dropping in a new state invalidates in-flight gradient bookkeeping
