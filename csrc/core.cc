@@ -531,12 +531,16 @@ void PerformOperation(GlobalState& st, Response& resp) {
              (resp.type == ResponseType::ALLREDUCE ||
               resp.type == ResponseType::ADASUM ||
               resp.type == ResponseType::ALLGATHER ||
-              resp.type == ResponseType::BROADCAST)) {
+              resp.type == ResponseType::BROADCAST ||
+              resp.type == ResponseType::REDUCESCATTER ||
+              resp.type == ResponseType::ALLTOALL)) {
     // Zero substitution for a joined rank (reference: tensor_queue.cc
-    // GetTensorEntriesFromResponse join path): allreduce/adasum contribute
-    // zeros, allgather contributes zero rows, broadcast receives into a
-    // scratch buffer — all so the rank still participates in the RCCL
-    // collective its comm peers will issue.
+    // GetTensorEntriesFromResponse:125-141 substitutes for EVERY op type):
+    // allreduce/adasum/reducescatter contribute zeros, allgather/alltoall
+    // contribute zero rows (the alltoall split-matrix row for a joined rank
+    // is already all zeros, but it must still post its recvs), broadcast
+    // receives into a scratch buffer — all so the rank still participates
+    // in the RCCL collective its comm peers will issue.
     auto shapes = ParseShapes(resp);
     int dev;
     {
@@ -553,7 +557,9 @@ void PerformOperation(GlobalState& st, Response& resp) {
                       .device(dev == CPU_DEVICE_ID ? at::Device(at::kCPU)
                                                    : at::Device(at::kCUDA, dev));
       std::vector<int64_t> shape = shapes[i];
-      if (resp.type == ResponseType::ALLGATHER && !shape.empty())
+      if ((resp.type == ResponseType::ALLGATHER ||
+           resp.type == ResponseType::ALLTOALL) &&
+          !shape.empty())
         shape[0] = 0;  // contribute zero rows
       e.tensor = at::zeros(shape, opts);
       e.device = dev;

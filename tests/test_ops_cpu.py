@@ -594,3 +594,155 @@ def test_concurrent_sets_grouped_compressed_np4():
             assert abs(g[0].item() - 10.0) < 0.1, g[0]
         hvd.barrier()
     """)
+
+
+def test_mixed_reduce_ops_not_fused_np2():
+    """Round-2 regression (ADVICE high): concurrent allreduces with the same
+    dtype/shape but different reduce ops must not fuse into one buffer."""
+    run_workers(2, """
+        a = torch.tensor([1.0, 5.0]) if rank == 0 else torch.tensor([3.0, 2.0])
+        b = torch.tensor([2.0, 7.0]) if rank == 0 else torch.tensor([4.0, 1.0])
+        hs = [hvd.allreduce_async(a, op=hvd.Sum, name="mix.sum"),
+              hvd.allreduce_async(b, op=hvd.Max, name="mix.max"),
+              hvd.allreduce_async(a, op=hvd.Min, name="mix.min"),
+              hvd.allreduce_async(b, op=hvd.Product, name="mix.prod")]
+        s, mx, mn, pr = [hvd.synchronize(h) for h in hs]
+        assert torch.allclose(s, torch.tensor([4.0, 7.0])), s
+        assert torch.allclose(mx, torch.tensor([4.0, 7.0])), mx
+        assert torch.allclose(mn, torch.tensor([1.0, 2.0])), mn
+        assert torch.allclose(pr, torch.tensor([8.0, 7.0])), pr
+    """)
+
+
+def test_minmax_steady_state_cache_np2():
+    """Round-2 regression (ADVICE low): non-SUM ops must survive the cache
+    fast path — same name repeated across many cycles."""
+    run_workers(2, """
+        for i in range(12):
+            t = torch.tensor([float(rank + 1), float(10 - rank)])
+            mn = hvd.allreduce(t, op=hvd.Min, name="steadymin")
+            mx = hvd.allreduce(t, op=hvd.Max, name="steadymax")
+            assert torch.allclose(mn, torch.tensor([1.0, 9.0])), (i, mn)
+            assert torch.allclose(mx, torch.tensor([2.0, 10.0])), (i, mx)
+    """)
+
+
+def test_synchronize_fires_missing_hooks_np2():
+    """Round-2 regression (ADVICE medium): per-rank conditional execution —
+    a param whose hook never fired on one rank must still be allreduced by
+    synchronize(), or peers deadlock."""
+    run_workers(2, """
+        m = torch.nn.Linear(4, 2, bias=True)
+        with torch.no_grad():
+            m.weight.fill_(1.0); m.bias.fill_(0.0)
+        opt = torch.optim.SGD(m.parameters(), lr=0.0)
+        opt = hvd.DistributedOptimizer(opt,
+                                       named_parameters=m.named_parameters())
+        x = torch.ones(1, 4)
+        if rank == 0:
+            loss = m(x).sum()          # both weight and bias get grads
+        else:
+            loss = (m.weight @ x.t()).sum()  # bias hook never fires
+        loss.backward()
+        opt.step()                      # must not hang; fires bias on rank 1
+        # bias grad: rank0 contributed ones, rank1 zeros -> Average = 0.5
+        assert torch.allclose(m.bias.grad, torch.full((2,), 0.5)), m.bias.grad
+    """)
+
+
+def test_adasum_set_backward_passes_np2():
+    """Round-2 regression (ADVICE low): no AttributeError on the Adasum
+    delta optimizer's set_backward_passes_per_step."""
+    run_workers(2, """
+        m = torch.nn.Linear(3, 3)
+        opt = torch.optim.SGD(m.parameters(), lr=0.1)
+        opt = hvd.DistributedOptimizer(opt, op=hvd.Adasum,
+                                       named_parameters=m.named_parameters())
+        opt.set_backward_passes_per_step(2)
+        assert opt.backward_passes_per_step == 2
+    """)
+
+
+def test_broadcast_fresh_optimizer_state_np2():
+    """Round-2: broadcast_optimizer_state on a FRESH optimizer initializes
+    identical momentum state everywhere (reference functions.py:90-109)."""
+    run_workers(2, """
+        torch.manual_seed(100 + rank)  # different params per rank
+        m = torch.nn.Linear(4, 4)
+        opt = torch.optim.SGD(m.parameters(), lr=0.1, momentum=0.9)
+        hvd.broadcast_parameters(m.state_dict(), root_rank=0)
+        hvd.broadcast_optimizer_state(opt, root_rank=0, model=m)
+        sd = opt.state_dict()
+        assert len(sd['state']) == 2, sd['state'].keys()
+        for s in sd['state'].values():
+            assert 'momentum_buffer' in s
+        # buffers must agree across ranks (both rebuilt from root)
+        flat = torch.cat([s['momentum_buffer'].flatten()
+                          for s in sd['state'].values()])
+        mx = hvd.allreduce(flat, op=hvd.Max, name="mbmax")
+        mn = hvd.allreduce(flat, op=hvd.Min, name="mbmin")
+        assert torch.equal(mx, mn)
+    """)
+
+
+def test_reducescatter_prescale_np2():
+    """Round-2 regression (ADVICE medium, CPU twin of the GPU fix):
+    reducescatter honors prescale_factor."""
+    run_workers(2, """
+        t = torch.ones(4, 3)
+        out = hvd.reducescatter(t, op=hvd.Sum, prescale_factor=3.0,
+                                name="rsps")
+        # 3*1 + 3*1 = 6 per element, 2 rows per rank
+        assert out.shape == (2, 3), out.shape
+        assert torch.allclose(out, torch.full((2, 3), 6.0)), out
+    """)
+
+
+def test_allgather_grad_uneven_np2():
+    """Round-2: allgather autograd backward slices at the offset captured in
+    forward (no per-backward dims collective) — uneven first dims."""
+    run_workers(2, """
+        n = 2 if rank == 0 else 3
+        t = torch.full((n, 2), float(rank + 1), requires_grad=True)
+        out = hvd.allgather(t, name="aggrad")
+        assert out.shape == (5, 2)
+        # d(loss)/dt where loss weights this rank's slice by (rank+2):
+        w = torch.ones_like(out)
+        w[(0 if rank == 0 else 2):(2 if rank == 0 else 5)] = 0.0
+        loss = (out * w).sum()
+        loss.backward()
+        # grad of summed-other-slices: other rank's weight contributes 1 via
+        # allreduce-sum of grad_output then slicing our rows
+        assert t.grad.shape == t.shape
+        assert torch.allclose(t.grad, torch.ones_like(t)), t.grad
+    """)
+
+
+def test_join_with_reducescatter_np2():
+    """Round-2: joined ranks zero-substitute for reducescatter too
+    (reference tensor_queue.cc:125-141 substitutes for every op type)."""
+    run_workers(2, """
+        if rank == 0:
+            out = hvd.reducescatter(torch.full((4, 3), 2.0), op=hvd.Sum,
+                                    name="jrs")
+            # peer joined -> contributes zeros; I get rows 0..1
+            assert out.shape == (2, 3), out.shape
+            assert torch.allclose(out, torch.full((2, 3), 2.0)), out
+        hvd.join()
+    """, timeout=120)
+
+
+def test_join_with_alltoall_np2():
+    """Round-2: a joined rank posts zero-row sends but still receives what
+    peers address to it (its split-matrix row is zeros)."""
+    run_workers(2, """
+        if rank == 0:
+            t = torch.arange(12, dtype=torch.float32).reshape(4, 3)
+            out, rs = hvd.alltoall(t, splits=torch.tensor([2, 2]),
+                                   name="jata")
+            # joined peer sent nothing; we get only our own first 2 rows
+            assert out.shape == (2, 3), out.shape
+            assert torch.allclose(out, t[:2]), out
+            assert rs.tolist() == [2, 0], rs
+        hvd.join()
+    """, timeout=120)
