@@ -614,7 +614,10 @@ void PerformOperation(GlobalState& st, Response& resp) {
       if (tl) tl->OpEnd(resp);
     }
   } catch (const std::exception& ex) {
-    FailEntries(entries, Status::UnknownError(ex.what()));
+    // a RCCL call failing because comms were aborted is the elastic-
+    // recoverable signal, not an internal bug
+    FailEntries(entries, gpu::CommsFailed() ? Status::Aborted(ex.what())
+                                            : Status::UnknownError(ex.what()));
     if (tl) tl->OpEnd(resp);
   }
 }
@@ -623,6 +626,11 @@ void Abort(GlobalState& st, const std::string& why) {
   st.aborted = true;
   st.abort_reason = why;
   HVD_LOG(ERROR, "background loop aborted: %s", why.c_str());
+  // Tear down the GPU data plane FIRST: a peer that died mid-collective
+  // leaves RCCL kernels hung on the comm stream; ncclCommAbort unblocks
+  // them so the finalizer can fail their handles (and so WaitAllPending in
+  // the shutdown path terminates).
+  gpu::AbortComms(why);
   auto s = Status::Aborted(why);
   st.queue.FailAll(s);
   st.handles.FailAll(s);

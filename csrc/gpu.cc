@@ -98,6 +98,7 @@ struct PendingOp {
   int device = -1;
   int64_t start_us = 0;
   std::string activity;
+  ncclComm_t comm = nullptr;  // for the async-error watchdog
 };
 
 struct Finalizer {
@@ -112,6 +113,20 @@ struct Finalizer {
 std::mutex g_ctx_mu;
 std::unordered_map<int, DeviceCtx*> g_ctx;
 std::atomic<bool> g_rccl_used{false};
+// Set once any communicator is aborted (async RCCL error or TCP peer loss).
+// Every subsequent GPU op fails fast with ABORTED — the signal elastic
+// recovery converts into HorovodInternalError (reference:
+// nccl_operations.cc:56-147 commDestroyOrAbort + elastic_restart_).
+std::atomic<bool> g_comm_failed{false};
+// guards every DeviceCtx::comms map (bg thread inserts, finalizer aborts)
+std::mutex g_comms_mu;
+
+bool CommHasAsyncError(ncclComm_t comm) {
+  if (!comm) return false;
+  ncclResult_t async_err = ncclSuccess;
+  if (ncclCommGetAsyncError(comm, &async_err) != ncclSuccess) return true;
+  return async_err != ncclSuccess && async_err != ncclInProgress;
+}
 
 // hipEvent pool (reference: GPUContext event pool with prepopulation,
 // hip_operations.cc:5-80) — create/destroy cost ~2 us each adds up at 161
@@ -141,6 +156,8 @@ void ReleaseEvent(hipEvent_t ev) {
     (void)hipEventDestroy(ev);
 }
 
+void AbortAllCommsLocked(const char* why);
+
 void FinalizerLoop() {
   while (true) {
     PendingOp op;
@@ -160,17 +177,52 @@ void FinalizerLoop() {
     // locks (measured: +150% step time under 161-tensor gradient flow);
     // 50 us sleeps make the query rate negligible while keeping completion
     // latency far below a bucket's comm time.
+    //
+    // Watchdog (reference: nccl_operations.cc AsyncErrorCheck): every ~10 ms
+    // of waiting, poll ncclCommGetAsyncError on this op's comm and the
+    // global failure flag; on error, ncclCommAbort every comm so the hung
+    // collective unblocks, then fail the entries with ABORTED so user-side
+    // synchronize() raises HorovodInternalError instead of hanging.
     hipError_t e;
-    while ((e = hipEventQuery(op.done_event)) == hipErrorNotReady)
+    int spins = 0;
+    bool comm_dead = false;
+    auto fail_deadline = std::chrono::steady_clock::time_point::max();
+    while ((e = hipEventQuery(op.done_event)) == hipErrorNotReady) {
       std::this_thread::sleep_for(std::chrono::microseconds(50));
-    Status s = e == hipSuccess
-                   ? Status::OK()
-                   : Status::UnknownError(std::string("hipEventSynchronize: ") +
-                                          hipGetErrorString(e));
+      if (++spins % 200 != 0) continue;  // async checks every ~10 ms
+      auto now = std::chrono::steady_clock::now();
+      if (comm_dead) {
+        if (now > fail_deadline) break;  // abort didn't unblock the stream
+        continue;
+      }
+      if (g_comm_failed || CommHasAsyncError(op.comm)) {
+        {
+          std::lock_guard<std::mutex> g(g_comms_mu);
+          AbortAllCommsLocked("RCCL async error detected by finalizer");
+        }
+        comm_dead = true;
+        // give ncclCommAbort 2 s to terminate the enqueued kernel cleanly
+        fail_deadline = now + std::chrono::seconds(2);
+      }
+    }
+    Status s;
+    if (comm_dead) {
+      s = Status::Aborted(
+          "RCCL communicator aborted (peer failure or async error)");
+    } else if (e == hipSuccess) {
+      s = Status::OK();
+    } else {
+      s = Status::UnknownError(std::string("hipEventSynchronize: ") +
+                               hipGetErrorString(e));
+    }
     for (auto& entry : op.entries)
       if (entry.callback) entry.callback(s, entry);
-    ReleaseEvent(op.done_event);
-    for (auto ev : op.ready_events) ReleaseEvent(ev);
+    if (e != hipErrorNotReady) {
+      // only recycle events that actually signaled; a wedged event must not
+      // re-enter the pool (it would poison a future op)
+      ReleaseEvent(op.done_event);
+      for (auto ev : op.ready_events) ReleaseEvent(ev);
+    }
     auto& st = State();
     auto tl = GetTimeline(st);
     if (tl && !op.entries.empty())
@@ -206,11 +258,25 @@ std::string ExchangeUniqueId(GlobalState& st, int leader,
   return st.comm.Bcast(idb);
 }
 
+// Abort every live communicator (called with g_comms_mu held).  After this,
+// all GPU collectives fail fast until the next elastic re-init.
+void AbortAllCommsLocked(const char* why) {
+  if (g_comm_failed.exchange(true)) return;  // once
+  HVD_LOG(ERROR, "aborting all RCCL comms: %s", why);
+  for (auto& kv : g_ctx) {
+    for (auto& ck : kv.second->comms) (void)ncclCommAbort(ck.second);
+    kv.second->comms.clear();
+  }
+}
+
 // Lazy RCCL communicator for a process set (reference: nccl_operations.cc
 // 87-131 — id bcast via the controller, lock-step on every global rank).
 ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
-  auto it = ctx.comms.find(set_id);
-  if (it != ctx.comms.end()) return it->second;
+  {
+    std::lock_guard<std::mutex> g(g_comms_mu);
+    auto it = ctx.comms.find(set_id);
+    if (it != ctx.comms.end()) return it->second;
+  }
   auto& set = st.controller->process_set(set_id);
   int leader = set.ranks.empty() ? 0 : set.ranks[0];
 
@@ -229,7 +295,10 @@ ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
   ncclComm_t comm = nullptr;
   RCCL_CHECK(ncclCommInitRank(&comm, (int)set.ranks.size(), id,
                               set.local_index(st.rank)));
-  ctx.comms[set_id] = comm;
+  {
+    std::lock_guard<std::mutex> g(g_comms_mu);
+    ctx.comms[set_id] = comm;
+  }
   g_rccl_used = true;
   HVD_LOG(INFO, "RCCL comm ready: process set %d, %d ranks, device %d",
           (int)set_id, (int)set.ranks.size(), ctx.device);
@@ -297,7 +366,7 @@ void WaitReadyEvents(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
 
 void Finalize(DeviceCtx& ctx, std::vector<TensorTableEntry> entries,
               std::vector<hipEvent_t> ready, const char* activity,
-              int64_t start_us) {
+              int64_t start_us, ncclComm_t comm = nullptr) {
   PendingOp op;
   op.done_event = AcquireEvent();
   HIP_CHECK(hipEventRecord(op.done_event, ctx.stream.stream()));
@@ -306,6 +375,7 @@ void Finalize(DeviceCtx& ctx, std::vector<TensorTableEntry> entries,
   op.device = ctx.device;
   op.activity = activity;
   op.start_us = start_us;
+  op.comm = comm;
   EnsureFinalizer();
   {
     std::lock_guard<std::mutex> g(g_finalizer.mu);
@@ -468,6 +538,15 @@ bool RcclUsed() { return g_rccl_used; }
 
 void Execute(GlobalState& st, Response& resp,
              std::vector<TensorTableEntry>& entries) {
+  if (g_comm_failed) {
+    // fail fast until elastic re-init replaces the comms — issuing RCCL
+    // calls on an aborted comm would hang or corrupt the stream
+    Status s = Status::Aborted(
+        "RCCL communicator aborted (peer failure or async error)");
+    for (auto& e : entries)
+      if (e.callback) e.callback(s, e);
+    return;
+  }
   auto& set = st.controller->process_set(resp.process_set_id);
   if (!g_bootstrapped.count(resp.process_set_id)) {
     // Every global rank reaches this point on the set's first GPU response
@@ -719,8 +798,15 @@ void Execute(GlobalState& st, Response& resp,
   }
 
   if (roctx_on) roctxRangePop();
-  Finalize(ctx, std::move(entries), std::move(ready), activity, t_start);
+  Finalize(ctx, std::move(entries), std::move(ready), activity, t_start, comm);
 }
+
+void AbortComms(const std::string& why) {
+  std::lock_guard<std::mutex> g(g_comms_mu);
+  AbortAllCommsLocked(why.c_str());
+}
+
+bool CommsFailed() { return g_comm_failed; }
 
 void AdasumCombine(std::vector<at::Tensor>& a, std::vector<at::Tensor>& b) {
   if (a.empty()) return;
@@ -884,12 +970,15 @@ void Shutdown() {
   g_finalizer.stop = false;
   {
     std::lock_guard<std::mutex> g(g_ctx_mu);
+    std::lock_guard<std::mutex> g2(g_comms_mu);
     for (auto& kv : g_ctx) {
+      // comms already aborted (and map cleared) when g_comm_failed
       for (auto& ck : kv.second->comms) ncclCommDestroy(ck.second);
       delete kv.second;
     }
     g_ctx.clear();
   }
+  g_comm_failed = false;  // elastic re-init starts clean
   g_bootstrapped.clear();
   std::lock_guard<std::mutex> g(g_event_mu);
   for (auto ev : g_event_pool) (void)hipEventDestroy(ev);

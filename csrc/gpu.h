@@ -33,6 +33,14 @@ void Execute(GlobalState& st, Response& resp,
 void WaitAllPending();
 void Shutdown();
 
+// Abort every live RCCL communicator (TCP peer loss, async RCCL error, or
+// stall shutdown).  Idempotent; unblocks hung collectives so pending
+// handles fail with ABORTED (-> HorovodInternalError) instead of hanging.
+// Reference: nccl_operations.cc:56-147 commDestroyOrAbort.
+void AbortComms(const std::string& why);
+// True once AbortComms ran (cleared by Shutdown for elastic re-init).
+bool CommsFailed();
+
 // True once a RCCL communicator has been created (used by tests to assert
 // the native path ran).
 bool RcclUsed();

@@ -290,3 +290,63 @@ def test_multi_grouped_and_concurrent_sets():
             for o in o2:
                 assert torch.allclose(o, torch.full_like(o, 2.0 * tri))
     """, timeout=420)
+
+
+@requires_multi_gpu
+def test_multi_peer_death_aborts_rccl():
+    """Kill one rank mid-allreduce stream: survivors must raise
+    HorovodInternalError within a bound, not hang in RCCL (round-2 watchdog:
+    TCP peer loss -> ncclCommAbort -> pending handles fail ABORTED)."""
+    import os
+    import subprocess
+    import sys
+    import time
+    from horovod_amd.runner.launch import find_free_port, slot_env
+    from tests.parallel_util import REPO
+
+    np_ = min(NGPU, 4)
+    victim_body = (
+        "import torch, horovod_amd.torch as hvd, os\n"
+        "hvd.init(); torch.cuda.set_device(hvd.local_rank())\n"
+        "d = torch.device('cuda', hvd.local_rank())\n"
+        "hvd.allreduce(torch.ones(1024, device=d), name='fd0')\n"
+        "os._exit(9)  # die with a comm established\n"
+    )
+    survivor_body = (
+        "import torch, horovod_amd.torch as hvd\n"
+        "from horovod_amd.common.exceptions import HorovodInternalError\n"
+        "hvd.init(); torch.cuda.set_device(hvd.local_rank())\n"
+        "d = torch.device('cuda', hvd.local_rank())\n"
+        "hvd.allreduce(torch.ones(1024, device=d), name='fd0')\n"
+        "try:\n"
+        "    for i in range(1, 50):\n"
+        "        hvd.allreduce(torch.ones(1 << 20, device=d), name=f'fd{i}')\n"
+        "    print('UNEXPECTED_SUCCESS')\n"
+        "except (HorovodInternalError, RuntimeError) as e:\n"
+        "    print('GOT_ERROR', type(e).__name__)\n"
+    )
+    port = find_free_port()
+    procs = []
+    for r in range(np_):
+        env = slot_env(r, np_, r, np_, 0, 1, "127.0.0.1", port)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        env["HOROVOD_SHUTDOWN_GRACE_SECONDS"] = "2"
+        body = victim_body if r == np_ - 1 else survivor_body
+        procs.append(subprocess.Popen([sys.executable, "-c", body], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, text=True))
+    t0 = time.time()
+    outs = []
+    for r, p in enumerate(procs):
+        try:
+            out, _ = p.communicate(timeout=180)
+        except subprocess.TimeoutExpired:
+            for q in procs:
+                q.kill()
+            raise AssertionError(
+                f"rank {r} hung >180s after peer death (watchdog failed)")
+        outs.append(out)
+    elapsed = time.time() - t0
+    for r in range(np_ - 1):
+        assert "GOT_ERROR" in outs[r], (r, outs[r][-500:])
+    assert elapsed < 150, f"survivors took {elapsed:.0f}s to fail"
