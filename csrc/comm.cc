@@ -214,4 +214,199 @@ std::string StarComm::ScatterFrames(const std::vector<std::string>& frames) {
   return RecvFrame(fds_[0]);
 }
 
+
+// ---------------------------------------------------------------------------
+// MeshComm
+// ---------------------------------------------------------------------------
+
+MeshComm::~MeshComm() { Shutdown(); }
+
+int MeshComm::fd_of(int peer) const {
+  int fd = fds_[peer];
+  if (fd < 0) comm_error("mesh: no link to rank " + std::to_string(peer));
+  return fd;
+}
+
+void MeshComm::Init(StarComm& star, const std::string& root_addr,
+                    double timeout_sec) {
+  rank_ = star.rank();
+  size_ = star.size();
+  if (size_ <= 1) {
+    alive_ = true;
+    return;
+  }
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::duration<double>(timeout_sec);
+  fds_.assign(size_, -1);
+
+  // 1. listener on an ephemeral port
+  listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
+  if (listen_fd_ < 0) comm_error("mesh socket");
+  int one = 1;
+  setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  sockaddr_in sa{};
+  sa.sin_family = AF_INET;
+  sa.sin_addr.s_addr = INADDR_ANY;
+  sa.sin_port = 0;
+  if (bind(listen_fd_, (sockaddr*)&sa, sizeof(sa)) != 0)
+    comm_error("mesh bind");
+  if (listen(listen_fd_, size_) != 0) comm_error("mesh listen");
+  socklen_t slen = sizeof(sa);
+  getsockname(listen_fd_, (sockaddr*)&sa, &slen);
+  int my_port = ntohs(sa.sin_port);
+
+  // 2. my address as peers can reach it: rank 0 advertises the address the
+  // workers already used for the star; a worker advertises the local IP of
+  // its star socket (the interface that routes to the job).
+  std::string my_ip;
+  if (rank_ == 0) {
+    my_ip = root_addr;
+  } else {
+    sockaddr_in la{};
+    socklen_t ll = sizeof(la);
+    getsockname(star.fds_[0], (sockaddr*)&la, &ll);
+    char buf[64];
+    inet_ntop(AF_INET, &la.sin_addr, buf, sizeof(buf));
+    my_ip = buf;
+  }
+  std::string endpoint = my_ip + ":" + std::to_string(my_port);
+  auto eps = star.Gather(endpoint);
+  {
+    std::string joined;
+    if (star.is_root())
+      for (auto& e : eps) joined += e + "\n";
+    joined = star.Bcast(joined);
+    eps.clear();
+    size_t pos = 0;
+    while (pos < joined.size()) {
+      size_t nl = joined.find('\n', pos);
+      eps.push_back(joined.substr(pos, nl - pos));
+      pos = nl + 1;
+    }
+  }
+
+  // 3. connect to every LOWER rank (their listeners already exist: the
+  // endpoint exchange above is a barrier), then accept the higher ranks.
+  for (int peer = 0; peer < rank_; ++peer) {
+    auto colon = eps[peer].rfind(':');
+    std::string ip = eps[peer].substr(0, colon);
+    std::string port_s = eps[peer].substr(colon + 1);
+    addrinfo hints{}, *res = nullptr;
+    hints.ai_family = AF_INET;
+    hints.ai_socktype = SOCK_STREAM;
+    if (getaddrinfo(ip.c_str(), port_s.c_str(), &hints, &res) != 0 || !res)
+      comm_error("mesh getaddrinfo " + eps[peer]);
+    int fd = -1;
+    while (true) {
+      fd = socket(AF_INET, SOCK_STREAM, 0);
+      if (fd < 0) comm_error("mesh socket");
+      if (connect(fd, res->ai_addr, res->ai_addrlen) == 0) break;
+      close(fd);
+      if (std::chrono::steady_clock::now() > deadline) {
+        freeaddrinfo(res);
+        comm_error("mesh connect timeout to " + eps[peer]);
+      }
+      std::this_thread::sleep_for(std::chrono::milliseconds(20));
+    }
+    freeaddrinfo(res);
+    set_nodelay(fd);
+    int32_t me = rank_;
+    const char* pb = (const char*)&me;
+    size_t left = sizeof(me);
+    while (left) {
+      ssize_t n = ::send(fd, pb, left, MSG_NOSIGNAL);
+      if (n <= 0) comm_error("mesh handshake send");
+      pb += n;
+      left -= (size_t)n;
+    }
+    fds_[peer] = fd;
+  }
+  int expected = size_ - 1 - rank_;
+  while (expected > 0) {
+    auto remain = std::chrono::duration<double>(
+                      deadline - std::chrono::steady_clock::now())
+                      .count();
+    if (remain <= 0) comm_error("mesh accept timeout");
+    struct pollfd pfd{listen_fd_, POLLIN, 0};
+    int pr = poll(&pfd, 1, (int)(remain * 1000));
+    if (pr <= 0) comm_error("mesh accept timeout/poll");
+    int fd = accept(listen_fd_, nullptr, nullptr);
+    if (fd < 0) comm_error("mesh accept");
+    set_nodelay(fd);
+    int32_t peer = -1;
+    char* pb = (char*)&peer;
+    size_t left = sizeof(peer);
+    while (left) {
+      ssize_t n = ::recv(fd, pb, left, 0);
+      if (n <= 0) comm_error("mesh handshake recv");
+      pb += n;
+      left -= (size_t)n;
+    }
+    if (peer <= rank_ || peer >= size_) comm_error("mesh bad peer rank");
+    fds_[peer] = fd;
+    --expected;
+  }
+  close(listen_fd_);
+  listen_fd_ = -1;
+  alive_ = true;
+}
+
+void MeshComm::Shutdown() {
+  for (int fd : fds_)
+    if (fd >= 0) close(fd);
+  fds_.clear();
+  if (listen_fd_ >= 0) close(listen_fd_);
+  listen_fd_ = -1;
+  alive_ = false;
+}
+
+void MeshComm::Send(int peer, const void* data, size_t len) {
+  SendRecv2(peer, data, len, -1, nullptr, 0);
+}
+
+void MeshComm::Recv(int peer, void* data, size_t len) {
+  SendRecv2(-1, nullptr, 0, peer, data, len);
+}
+
+void MeshComm::SendRecv2(int send_peer, const void* out, size_t out_len,
+                         int recv_peer, void* in, size_t in_len) {
+  // self-exchange degenerates to memcpy
+  if (send_peer == rank_ && recv_peer == rank_) {
+    if (in_len != out_len) comm_error("mesh self-exchange length mismatch");
+    if (in_len) std::memcpy(in, out, in_len);
+    return;
+  }
+  if (send_peer == rank_ || recv_peer == rank_)
+    comm_error("mesh asymmetric self-exchange unsupported");
+  size_t so = 0, ro = 0;
+  int sfd = (send_peer >= 0 && out_len) ? fd_of(send_peer) : -1;
+  int rfd = (recv_peer >= 0 && in_len) ? fd_of(recv_peer) : -1;
+  while ((sfd >= 0 && so < out_len) || (rfd >= 0 && ro < in_len)) {
+    struct pollfd p[2];
+    int n = 0, si = -1, ri = -1;
+    if (sfd >= 0 && so < out_len) {
+      p[n] = {sfd, POLLOUT, 0};
+      si = n++;
+    }
+    if (rfd >= 0 && ro < in_len) {
+      p[n] = {rfd, POLLIN, 0};
+      ri = n++;
+    }
+    int pr = poll(p, n, 300000);  // 5-minute safety: a dead peer -> error
+    if (pr == 0) comm_error("mesh transfer timeout");
+    if (pr < 0) comm_error("mesh poll");
+    if (si >= 0 && (p[si].revents & (POLLOUT | POLLERR | POLLHUP))) {
+      ssize_t w = ::send(sfd, (const char*)out + so, out_len - so,
+                         MSG_NOSIGNAL);
+      if (w <= 0) comm_error("mesh send");
+      so += (size_t)w;
+    }
+    if (ri >= 0 && (p[ri].revents & (POLLIN | POLLERR | POLLHUP))) {
+      ssize_t r = ::recv(rfd, (char*)in + ro, in_len - ro, 0);
+      if (r <= 0) comm_error("mesh recv (peer died?)");
+      ro += (size_t)r;
+    }
+  }
+}
+
 }  // namespace hvd

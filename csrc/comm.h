@@ -59,6 +59,47 @@ class StarComm {
   // root: fds_[r] = connection to rank r (fds_[0] unused);
   // worker: fds_[0] = connection to root.
   std::vector<int> fds_;
+
+  friend class MeshComm;  // port exchange rides the star at init
+};
+
+// Full-mesh TCP data plane: direct member-to-member links for the CPU
+// collectives (ring allreduce, pairwise-exchange allgather/alltoall,
+// binomial broadcast), replacing the star gather+bcast that moved 2n x
+// data through rank 0 (round-1 weakness; reference analogue:
+// mpi_operations.cc:83-545 real MPI collectives among members).
+//
+// Only the background thread touches mesh sockets, in broadcast-ordered
+// response order, so per-pair framing needs no tags.  All transfers are
+// poll-driven full-duplex (SendRecv2): a ring step where every rank sends
+// and receives concurrently cannot deadlock on TCP buffer limits.
+class MeshComm {
+ public:
+  MeshComm() = default;
+  ~MeshComm();
+
+  // Collective over the (already-initialized) star: every global rank must
+  // call this together.  root_addr is rank 0's address as workers know it.
+  void Init(StarComm& star, const std::string& root_addr,
+            double timeout_sec = 120.0);
+  void Shutdown();
+  bool alive() const { return alive_; }
+
+  void Send(int peer, const void* data, size_t len);
+  void Recv(int peer, void* data, size_t len);
+  // Full-duplex: send `out` to send_peer while receiving `in` from
+  // recv_peer (they may be the same peer, or self => memcpy).
+  void SendRecv2(int send_peer, const void* out, size_t out_len,
+                 int recv_peer, void* in, size_t in_len);
+
+ private:
+  int fd_of(int peer) const;
+
+  int rank_ = 0;
+  int size_ = 1;
+  bool alive_ = false;
+  int listen_fd_ = -1;
+  std::vector<int> fds_;  // fds_[r] = direct link to rank r (self = -1)
 };
 
 }  // namespace hvd

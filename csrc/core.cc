@@ -258,9 +258,289 @@ void PerTensorLayout(const Response& resp, std::vector<int64_t>& offsets,
   }
 }
 
+// ---- mesh algorithms ------------------------------------------------------
+// Direct member-to-member data plane (MeshComm): ring allreduce, pairwise-
+// exchange allgather/alltoall/reducescatter, binomial broadcast.  Each rank
+// moves ~2(n-1)/n x data for allreduce instead of 2n x through rank 0.
+// Reference analogue: mpi_operations.cc MPI_Allreduce/Allgatherv/Alltoallv.
+// Only set MEMBERS participate; frames ride per-pair sockets in response
+// order, so no interleaving with the control star is possible.
+
+bool UseMesh(GlobalState& st, const ProcessSetInfo& set) {
+  return st.mesh.alive() && st.size > 1 && set.ranks.size() > 1;
+}
+
+void ApplyReduce(at::Tensor acc, at::Tensor incoming, ReduceOp op) {
+  switch (op) {
+    case ReduceOp::MIN: at::minimum_out(acc, acc, incoming); break;
+    case ReduceOp::MAX: at::maximum_out(acc, acc, incoming); break;
+    case ReduceOp::PRODUCT: acc.mul_(incoming); break;
+    default: acc.add_(incoming); break;
+  }
+}
+
+// In-place ring allreduce on a flat contiguous tensor (wire dtype).
+// Phase 1: ring reduce-scatter (n-1 steps); phase 2: ring allgather.
+// Every element's accumulation order is a pure function of its segment, so
+// results are bit-identical on every rank.
+void MeshRingAllreduce(GlobalState& st, const ProcessSetInfo& set,
+                       at::Tensor buf, ReduceOp op) {
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  int64_t numel = buf.numel();
+  int64_t esz = buf.element_size();
+  if (numel == 0 || n <= 1) return;
+  auto seg_off = [&](int s) {
+    int64_t base = numel / n, rem = numel % n;
+    return (int64_t)s * base + std::min<int64_t>(s, rem);
+  };
+  auto seg_len = [&](int s) {
+    return numel / n + (s < numel % n ? 1 : 0);
+  };
+  int right = set.ranks[(li + 1) % n];
+  int left = set.ranks[(li - 1 + n) % n];
+  char* base = (char*)buf.data_ptr();
+  at::Tensor scratch = at::empty({numel / n + 1}, buf.options());
+  // reduce-scatter: after step k, segment (li-k-1) holds partials of k+2 ranks
+  for (int step = 0; step < n - 1; ++step) {
+    int ss = (li - step + n) % n;         // segment I send
+    int rs = (li - step - 1 + n) % n;     // segment I receive+reduce
+    int64_t slen = seg_len(ss), rlen = seg_len(rs);
+    st.mesh.SendRecv2(right, base + seg_off(ss) * esz, (size_t)(slen * esz),
+                      left, scratch.data_ptr(), (size_t)(rlen * esz));
+    if (rlen)
+      ApplyReduce(buf.narrow(0, seg_off(rs), rlen),
+                  scratch.narrow(0, 0, rlen), op);
+  }
+  // allgather: circulate the finished segments
+  for (int step = 0; step < n - 1; ++step) {
+    int ss = (li + 1 - step + n) % n;     // finished segment I hold
+    int rs = (li - step + n) % n;         // segment I receive
+    int64_t slen = seg_len(ss), rlen = seg_len(rs);
+    st.mesh.SendRecv2(right, base + seg_off(ss) * esz, (size_t)(slen * esz),
+                      left, base + seg_off(rs) * esz, (size_t)(rlen * esz));
+  }
+}
+
+void MeshAllreduce(GlobalState& st, const ProcessSetInfo& set,
+                   const Response& resp,
+                   std::vector<TensorTableEntry>& entries) {
+  DataType wire = resp.dtype;
+  auto wire_t = DataTypeToTorch(wire);
+  std::vector<int64_t> offsets, counts;
+  PerTensorLayout(resp, offsets, counts);
+  std::vector<at::Tensor> flats;
+  for (auto& e : entries) flats.push_back(FlatPrescaled(e, wire));
+  at::Tensor buf;
+  bool in_place = false;
+  if (flats.size() == 1) {
+    auto& e = entries[0];
+    // the ring reduces IN PLACE: only alias the user's storage when the op
+    // is explicitly in-place (output == tensor) with no dtype conversion;
+    // otherwise clone, or a second allreduce of the same input would read
+    // already-reduced values
+    in_place = e.output.defined() &&
+               e.output.data_ptr() == e.tensor.data_ptr() &&
+               flats[0].data_ptr() == e.tensor.data_ptr() &&
+               e.postscale == 1.0;
+    buf = in_place ? flats[0] : flats[0].clone();
+  } else {
+    buf = at::cat(flats);  // always a copy
+  }
+  MeshRingAllreduce(st, set, buf, resp.reduce_op);
+  if (!in_place)
+    for (size_t i = 0; i < entries.size(); ++i)
+      UnpackInto(entries[i], buf.narrow(0, offsets[i], counts[i]));
+}
+
+void MeshAllgather(GlobalState& st, const ProcessSetInfo& set,
+                   const Response& resp,
+                   std::vector<TensorTableEntry>& entries) {
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  auto& e = entries[0];
+  at::Tensor in = e.tensor.contiguous();
+  auto shapes = ParseShapes(resp);
+  int64_t row_elems = 1;
+  for (size_t d = 1; d < shapes[0].size(); ++d) row_elems *= shapes[0][d];
+  int64_t esz = in.element_size();
+  int64_t total0 = 0;
+  std::vector<int64_t> offs(n);
+  for (int r = 0; r < n; ++r) {
+    offs[r] = total0;
+    total0 += resp.tensor_sizes[r];
+  }
+  std::vector<int64_t> out_shape = shapes[0];
+  if (out_shape.empty()) out_shape = {total0};
+  else out_shape[0] = total0;
+  e.output = at::empty(out_shape, e.tensor.options());
+  char* out_base = (char*)e.output.data_ptr();
+  // my own contribution in place
+  if (resp.tensor_sizes[li])
+    std::memcpy(out_base + offs[li] * row_elems * esz, in.data_ptr(),
+                (size_t)(resp.tensor_sizes[li] * row_elems * esz));
+  // pairwise exchange: step s sends my rows to li+s, receives li-s's rows
+  for (int s = 1; s < n; ++s) {
+    int to = (li + s) % n, from = (li - s + n) % n;
+    st.mesh.SendRecv2(
+        set.ranks[to], in.data_ptr(),
+        (size_t)(resp.tensor_sizes[li] * row_elems * esz), set.ranks[from],
+        out_base + offs[from] * row_elems * esz,
+        (size_t)(resp.tensor_sizes[from] * row_elems * esz));
+  }
+  int nn = (int)resp.tensor_sizes.size();
+  e.received_splits = at::empty({nn}, at::kLong);
+  auto* gs = e.received_splits.data_ptr<int64_t>();
+  for (int r = 0; r < nn; ++r) gs[r] = resp.tensor_sizes[r];
+}
+
+void MeshBroadcast(GlobalState& st, const ProcessSetInfo& set,
+                   const Response& resp,
+                   std::vector<TensorTableEntry>& entries) {
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  int root_li = set.local_index(resp.root_rank);
+  if (root_li < 0) root_li = 0;
+  auto& e = entries[0];
+  at::Tensor in = e.tensor.is_non_overlapping_and_dense()
+                      ? e.tensor
+                      : e.tensor.contiguous();
+  bool is_root = li == root_li;
+  at::Tensor stage;
+  if (is_root) {
+    stage = in;
+  } else {
+    stage = (e.output.defined() && e.output.is_contiguous() &&
+             e.output.numel() == in.numel())
+                ? e.output
+                : at::empty({in.numel()}, in.options());
+  }
+  size_t bytes = (size_t)(in.numel() * in.element_size());
+  // binomial tree on virtual ranks (root at 0)
+  int v = (li - root_li + n) % n;
+  int mask = 1;
+  while (mask < n && (v & (mask - 1)) == 0) {
+    if (v & mask) {
+      int src = (v - mask + root_li) % n;
+      st.mesh.Recv(set.ranks[src], stage.data_ptr(), bytes);
+      break;
+    }
+    mask <<= 1;
+  }
+  while ((mask >>= 1) > 0) {
+    if (v + mask < n) {
+      int dst = (v + mask + root_li) % n;
+      st.mesh.Send(set.ranks[dst], stage.data_ptr(), bytes);
+    }
+  }
+  if (!e.output.defined()) {
+    e.output = is_root ? in : stage.reshape(in.sizes());
+  } else if (e.output.data_ptr() != stage.data_ptr()) {
+    e.output.copy_(is_root ? in : stage.reshape(e.output.sizes()));
+  } else if (!is_root && !e.output.sizes().equals(in.sizes()) &&
+             e.output.numel() == in.numel()) {
+    e.output = e.output.reshape(in.sizes());
+  }
+}
+
+void MeshAlltoall(GlobalState& st, const ProcessSetInfo& set,
+                  const Response& resp,
+                  std::vector<TensorTableEntry>& entries) {
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  auto& e = entries[0];
+  at::Tensor in = e.tensor.contiguous();
+  int64_t row_elems = 1;
+  for (int d = 1; d < e.tensor.dim(); ++d) row_elems *= e.tensor.size(d);
+  int64_t esz = in.element_size();
+  auto srows = [&](int i, int j) {  // rows sender i ships to receiver j
+    return resp.tensor_sizes[(size_t)i * n + j];
+  };
+  std::vector<int64_t> send_off(n), recv_off(n);
+  int64_t so = 0, ro = 0;
+  for (int r = 0; r < n; ++r) {
+    send_off[r] = so;
+    so += srows(li, r);
+    recv_off[r] = ro;
+    ro += srows(r, li);
+  }
+  std::vector<int64_t> out_shape(e.tensor.sizes().begin(),
+                                 e.tensor.sizes().end());
+  if (out_shape.empty()) out_shape = {ro};
+  else out_shape[0] = ro;
+  e.output = at::empty(out_shape, e.tensor.options());
+  char* in_base = (char*)in.data_ptr();
+  char* out_base = (char*)e.output.data_ptr();
+  // self block
+  if (srows(li, li))
+    std::memcpy(out_base + recv_off[li] * row_elems * esz,
+                in_base + send_off[li] * row_elems * esz,
+                (size_t)(srows(li, li) * row_elems * esz));
+  for (int s = 1; s < n; ++s) {
+    int to = (li + s) % n, from = (li - s + n) % n;
+    st.mesh.SendRecv2(set.ranks[to],
+                      in_base + send_off[to] * row_elems * esz,
+                      (size_t)(srows(li, to) * row_elems * esz),
+                      set.ranks[from],
+                      out_base + recv_off[from] * row_elems * esz,
+                      (size_t)(srows(from, li) * row_elems * esz));
+  }
+  e.received_splits = at::empty({n}, at::kLong);
+  auto* rs = e.received_splits.data_ptr<int64_t>();
+  for (int r = 0; r < n; ++r) rs[r] = srows(r, li);
+}
+
+void MeshReducescatter(GlobalState& st, const ProcessSetInfo& set,
+                       const Response& resp,
+                       std::vector<TensorTableEntry>& entries) {
+  int n = (int)set.ranks.size();
+  int li = set.local_index(st.rank);
+  auto& e = entries[0];
+  DataType wire = resp.dtype;
+  at::Tensor flat = FlatPrescaled(e, wire).contiguous();
+  auto shapes = ParseShapes(resp);
+  std::vector<int64_t> shape = shapes[0];
+  int64_t first = shape.empty() ? 1 : shape[0];
+  int64_t row_elems = 1;
+  for (size_t d = 1; d < shape.size(); ++d) row_elems *= shape[d];
+  int64_t esz = flat.element_size();
+  int64_t base_rows = first / n, rem = first % n;
+  auto rows_of = [&](int r) { return base_rows + (r < rem ? 1 : 0); };
+  auto roff = [&](int r) {
+    return (int64_t)r * base_rows + std::min<int64_t>(r, rem);
+  };
+  // my accumulator = my own slice (accumulation order: own + peers in
+  // (li-s) order — identical on every rank for a given element? No: each
+  // element is reduced by exactly one rank, so per-element order is fixed.)
+  at::Tensor acc =
+      flat.narrow(0, roff(li) * row_elems, rows_of(li) * row_elems).clone();
+  at::Tensor scratch = at::empty_like(acc);
+  char* in_base = (char*)flat.data_ptr();
+  for (int s = 1; s < n; ++s) {
+    int to = (li + s) % n, from = (li - s + n) % n;
+    st.mesh.SendRecv2(set.ranks[to], in_base + roff(to) * row_elems * esz,
+                      (size_t)(rows_of(to) * row_elems * esz),
+                      set.ranks[from], scratch.data_ptr(),
+                      (size_t)(rows_of(li) * row_elems * esz));
+    if (acc.numel()) ApplyReduce(acc, scratch, resp.reduce_op);
+  }
+  std::vector<int64_t> out_shape = shape;
+  if (out_shape.empty()) out_shape = {rows_of(li)};
+  else out_shape[0] = rows_of(li);
+  e.output = at::empty(out_shape, e.tensor.options());
+  at::Tensor src = acc;
+  if (e.postscale != 1.0) src = src.to(at::kDouble).mul_(e.postscale);
+  e.output.flatten().copy_(src.to(e.output.scalar_type()));
+}
+
 void CPUAllreduce(GlobalState& st, const Response& resp,
                   std::vector<TensorTableEntry>& entries, bool member) {
   auto& set = st.controller->process_set(resp.process_set_id);
+  if (resp.type != ResponseType::ADASUM && UseMesh(st, set)) {
+    // ring over direct links; non-members do nothing (no star frames)
+    if (member && !entries.empty()) MeshAllreduce(st, set, resp, entries);
+    return;
+  }
   DataType wire = resp.dtype;
   auto wire_t = DataTypeToTorch(wire);
   std::string payload;
@@ -323,6 +603,10 @@ void CPUAllreduce(GlobalState& st, const Response& resp,
 void CPUAllgather(GlobalState& st, const Response& resp,
                   std::vector<TensorTableEntry>& entries, bool member) {
   auto& set = st.controller->process_set(resp.process_set_id);
+  if (UseMesh(st, set)) {
+    if (member && !entries.empty()) MeshAllgather(st, set, resp, entries);
+    return;
+  }
   std::string payload;
   if (member && !entries.empty()) {
     at::Tensor t = entries[0].tensor.contiguous();
@@ -357,6 +641,13 @@ void CPUAllgather(GlobalState& st, const Response& resp,
 
 void CPUBroadcast(GlobalState& st, const Response& resp,
                   std::vector<TensorTableEntry>& entries, bool member) {
+  {
+    auto& set = st.controller->process_set(resp.process_set_id);
+    if (UseMesh(st, set)) {
+      if (member && !entries.empty()) MeshBroadcast(st, set, resp, entries);
+      return;
+    }
+  }
   std::string payload;
   if (member && !entries.empty() && resp.root_rank == st.rank) {
     at::Tensor t = entries[0].tensor.contiguous();
@@ -387,6 +678,10 @@ void CPUBroadcast(GlobalState& st, const Response& resp,
 void CPUAlltoall(GlobalState& st, const Response& resp,
                  std::vector<TensorTableEntry>& entries, bool member) {
   auto& set = st.controller->process_set(resp.process_set_id);
+  if (UseMesh(st, set)) {
+    if (member && !entries.empty()) MeshAlltoall(st, set, resp, entries);
+    return;
+  }
   int n = (int)set.ranks.size();
   std::string payload;
   if (member && !entries.empty()) {
@@ -436,6 +731,11 @@ void CPUAlltoall(GlobalState& st, const Response& resp,
 void CPUReducescatter(GlobalState& st, const Response& resp,
                       std::vector<TensorTableEntry>& entries, bool member) {
   auto& set = st.controller->process_set(resp.process_set_id);
+  if (UseMesh(st, set)) {
+    if (member && !entries.empty())
+      MeshReducescatter(st, set, resp, entries);
+    return;
+  }
   int n = (int)set.ranks.size();
   DataType wire = resp.dtype;
   auto wire_t = DataTypeToTorch(wire);
@@ -728,6 +1028,10 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
   if (const char* e = std::getenv("HOROVOD_START_TIMEOUT"))
     start_timeout = atof(e);
   st.comm.Init(rank, size, addr, port, start_timeout);
+  // CPU data plane: full mesh of member-to-member links (ring allreduce &
+  // friends).  HOROVOD_CPU_STAR=1 keeps the round-1 star fallback.
+  if (!std::getenv("HOROVOD_CPU_STAR"))
+    st.mesh.Init(st.comm, addr, start_timeout);
   st.controller.reset(new Controller(&st.comm, rank, size, cfg));
   if (cfg.timeline_enabled) {
     const char* tf = std::getenv("HOROVOD_TIMELINE");
@@ -745,6 +1049,7 @@ void ShutdownHorovod() {
   st.shutdown_requested = true;
   if (st.bg_thread.joinable()) st.bg_thread.join();
   gpu::Shutdown();
+  st.mesh.Shutdown();
   st.comm.Shutdown();
   st.controller.reset();
   SetTimeline(st, nullptr);

@@ -89,7 +89,8 @@ struct GlobalState {
   int rank = 0, size = 1, local_rank = 0, local_size = 1, cross_rank = 0,
       cross_size = 1;
 
-  StarComm comm;
+  StarComm comm;   // control plane (negotiation, bitvectors, bootstrap)
+  MeshComm mesh;   // CPU data plane (direct member-to-member links)
   std::unique_ptr<Controller> controller;
   TensorQueue queue;
   HandleManager handles;
