@@ -70,6 +70,24 @@ ncclRedOp_t ToNcclOp(ReduceOp op) {
   }
 }
 
+// One-shot allreduce shared-memory window for one (device, process set):
+// my coarse staging (slots x threshold) + fine-grained flag page, plus the
+// IPC-opened peer windows, set-local indexed.  See oneshot.hip for the
+// device protocol.
+struct OneshotCtx {
+  bool ready = false;
+  bool failed = false;
+  char* my_staging = nullptr;
+  unsigned long long* my_flags = nullptr;
+  void* peer_staging[kOneshotMaxRanks] = {};
+  void* peer_flags[kOneshotMaxRanks] = {};
+  std::vector<void*> opened;  // to hipIpcCloseMemHandle on shutdown
+  unsigned long long seq = 0;
+  int64_t slot_bytes = 0;
+  int nslots = 2;
+  int n = 0, li = -1;
+};
+
 struct DeviceCtx {
   int device = -1;
   // Dedicated comm stream from torch's pool.  NORMAL priority by default:
@@ -83,6 +101,7 @@ struct DeviceCtx {
   at::Tensor adasum_buffer;    // [set_size x fused] gather space for adasum
   at::Tensor dots_buffer;      // adasum per-tensor {dot,|a|2,|b|2} doubles
   std::unordered_map<int32_t, ncclComm_t> comms;  // process_set -> comm
+  std::unordered_map<int32_t, OneshotCtx> oneshot;  // process_set -> window
 
   explicit DeviceCtx(int dev)
       : device(dev),
@@ -266,6 +285,16 @@ void AbortAllCommsLocked(const char* why) {
   for (auto& kv : g_ctx) {
     for (auto& ck : kv.second->comms) (void)ncclCommAbort(ck.second);
     kv.second->comms.clear();
+    // raise the one-shot abort word (flags[3]) so spin-wait kernels exit
+    // instead of wedging the GPU on a dead peer (SDMA write proceeds even
+    // while compute is spinning)
+    for (auto& ok : kv.second->oneshot) {
+      if (ok.second.my_flags) {
+        unsigned long long one = 1;
+        (void)hipMemcpy(ok.second.my_flags + 3, &one, sizeof(one),
+                        hipMemcpyHostToDevice);
+      }
+    }
   }
 }
 
@@ -308,6 +337,189 @@ ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
 // Process sets whose RCCL comm bootstrap already ran on this rank (background
 // thread only — every rank passes the same first-response for a set).
 std::unordered_map<int32_t, bool> g_bootstrapped;
+// Same for the one-shot shared-memory windows.
+std::unordered_map<int32_t, bool> g_oneshot_bootstrapped;
+
+// ---- one-shot allreduce host side -----------------------------------------
+
+int64_t OneshotThreshold() {
+  static int64_t v = [] {
+    const char* e = std::getenv("HOROVOD_ONESHOT_THRESHOLD");
+    return e ? (int64_t)atoll(e) : (int64_t)(4 << 20);
+  }();
+  return v;
+}
+
+bool OneshotEnabled() {
+  static bool v = std::getenv("HOROVOD_ONESHOT_ALLREDUCE") != nullptr;
+  return v;
+}
+
+// Eligibility is a pure function of the RESPONSE (not the local entries), so
+// every global rank — member or relay — classifies identically and the
+// star-frame lockstep holds.
+bool OneshotEligible(const Response& resp, int n) {
+  if (!OneshotEnabled() || resp.type != ResponseType::ALLREDUCE) return false;
+  if (n < 2 || n > kOneshotMaxRanks) return false;
+  switch (resp.dtype) {
+    case DataType::HVD_FLOAT32:
+    case DataType::HVD_FLOAT64:
+    case DataType::HVD_FLOAT16:
+    case DataType::HVD_BFLOAT16: break;
+    default: return false;
+  }
+  switch (resp.reduce_op) {
+    case ReduceOp::SUM:
+    case ReduceOp::AVERAGE:
+    case ReduceOp::MIN:
+    case ReduceOp::MAX:
+    case ReduceOp::PRODUCT: break;
+    default: return false;
+  }
+  int64_t total = 0;
+  const int64_t* p = resp.tensor_shapes.data();
+  const int64_t* end = p + resp.tensor_shapes.size();
+  while (p < end) {
+    int64_t nd = *p++;
+    int64_t numel = 1;
+    for (int64_t i = 0; i < nd; ++i) numel *= *p++;
+    total += AlignedElems(numel);
+  }
+  return total * (int64_t)DataTypeSize(resp.dtype) <= OneshotThreshold();
+}
+
+int OneshotOpCode(ReduceOp op) {
+  switch (op) {
+    case ReduceOp::MIN: return 1;
+    case ReduceOp::MAX: return 2;
+    case ReduceOp::PRODUCT: return 3;
+    default: return 0;  // sum / average
+  }
+}
+
+// All-rank string exchange over the star (every global rank participates).
+std::vector<std::string> ExchangeAllGather(GlobalState& st,
+                                           const std::string& payload) {
+  auto gathered = st.comm.Gather(payload);
+  std::string joined;
+  if (st.comm.is_root()) {
+    for (auto& g : gathered) {
+      uint32_t len = (uint32_t)g.size();
+      joined.append((const char*)&len, sizeof(len));
+      joined += g;
+    }
+  }
+  joined = st.comm.Bcast(joined);
+  std::vector<std::string> out;
+  size_t pos = 0;
+  while (pos + sizeof(uint32_t) <= joined.size()) {
+    uint32_t len;
+    std::memcpy(&len, joined.data() + pos, sizeof(len));
+    pos += sizeof(len);
+    out.push_back(joined.substr(pos, len));
+    pos += len;
+  }
+  return out;
+}
+
+// Set up the shared-memory window for (ctx.device, set).  Lock-step: every
+// global rank calls this (members allocate+open, non-members relay frames).
+// Any failure anywhere degrades EVERY member to the RCCL path (a second
+// star round agrees on the outcome) — never a mixed algorithm choice.
+bool EnsureOneshot(GlobalState& st, DeviceCtx* ctx, int32_t set_id,
+                   bool member) {
+  OneshotCtx* os = nullptr;
+  std::string payload;
+  bool ok = true;
+  if (member && ctx) {
+    os = &ctx->oneshot[set_id];
+    auto& set = st.controller->process_set(set_id);
+    os->n = (int)set.ranks.size();
+    os->li = set.local_index(st.rank);
+    os->slot_bytes = OneshotThreshold();
+    c10::hip::HIPGuard guard(ctx->device);
+    hipIpcMemHandle_t hs{}, hf{};
+    do {
+      if (hipMalloc(&os->my_staging,
+                    (size_t)(os->nslots * os->slot_bytes)) != hipSuccess) {
+        ok = false;
+        break;
+      }
+      if (hipExtMallocWithFlags((void**)&os->my_flags, 4096,
+                                hipDeviceMallocFinegrained) != hipSuccess) {
+        ok = false;
+        break;
+      }
+      if (hipMemset(os->my_staging, 0,
+                    (size_t)(os->nslots * os->slot_bytes)) != hipSuccess ||
+          hipMemset(os->my_flags, 0, 4096) != hipSuccess) {
+        ok = false;
+        break;
+      }
+      if (hipIpcGetMemHandle(&hs, os->my_staging) != hipSuccess ||
+          hipIpcGetMemHandle(&hf, os->my_flags) != hipSuccess) {
+        ok = false;
+        break;
+      }
+    } while (false);
+    if (ok) {
+      payload.assign((const char*)&hs, sizeof(hs));
+      payload.append((const char*)&hf, sizeof(hf));
+    }
+  }
+  auto all = ExchangeAllGather(st, payload);
+  if (member && os && ok) {
+    auto& set = st.controller->process_set(set_id);
+    c10::hip::HIPGuard guard(ctx->device);
+    for (int r = 0; r < os->n && ok; ++r) {
+      int grank = set.ranks[r];
+      if (grank == st.rank) {
+        os->peer_staging[r] = os->my_staging;
+        os->peer_flags[r] = os->my_flags;
+        continue;
+      }
+      if (grank >= (int)all.size() ||
+          all[grank].size() != 2 * sizeof(hipIpcMemHandle_t)) {
+        ok = false;
+        break;
+      }
+      hipIpcMemHandle_t hs{}, hf{};
+      std::memcpy(&hs, all[grank].data(), sizeof(hs));
+      std::memcpy(&hf, all[grank].data() + sizeof(hs), sizeof(hf));
+      void *ps = nullptr, *pf = nullptr;
+      if (hipIpcOpenMemHandle(&ps, hs, hipIpcMemLazyEnablePeerAccess) !=
+              hipSuccess ||
+          hipIpcOpenMemHandle(&pf, hf, hipIpcMemLazyEnablePeerAccess) !=
+              hipSuccess) {
+        ok = false;
+        break;
+      }
+      os->peer_staging[r] = ps;
+      os->peer_flags[r] = pf;
+      os->opened.push_back(ps);
+      os->opened.push_back(pf);
+    }
+  }
+  // unanimous verdict round: one byte per rank ('1' ok / '0' failed)
+  auto verdicts = ExchangeAllGather(st, member ? std::string(ok ? "1" : "0")
+                                               : std::string("1"));
+  bool all_ok = true;
+  for (auto& v : verdicts)
+    if (v == "0") all_ok = false;
+  if (member && os) {
+    os->ready = ok && all_ok;
+    os->failed = !os->ready;
+    if (os->failed)
+      HVD_LOG(WARNING,
+              "one-shot allreduce bootstrap failed for set %d; falling back "
+              "to RCCL",
+              (int)set_id);
+    else
+      HVD_LOG(INFO, "one-shot xGMI allreduce ready: set %d, %d ranks",
+              (int)set_id, os->n);
+  }
+  return all_ok && (!member || ok);
+}
 
 at::Tensor& FusionBuffer(DeviceCtx& ctx, int64_t bytes) {
   if (!ctx.fusion_buffer.defined() || ctx.fusion_buffer.numel() < bytes) {
@@ -548,17 +760,31 @@ void Execute(GlobalState& st, Response& resp,
     return;
   }
   auto& set = st.controller->process_set(resp.process_set_id);
-  if (!g_bootstrapped.count(resp.process_set_id)) {
-    // Every global rank reaches this point on the set's first GPU response
+  const int32_t sid = resp.process_set_id;
+  // Eligibility is response-derived, so members and relay-only ranks agree
+  // on which bootstrap (one-shot window vs RCCL comm) this response needs.
+  bool oneshot_resp = OneshotEligible(resp, (int)set.ranks.size());
+  auto rccl_bootstrap = [&] {
+    // Every global rank reaches this point on the set's first RCCL response
     // (responses are broadcast in a fixed order), so members can init the
     // RCCL comm while non-members relay the uniqueId frames in lock-step.
     if (!entries.empty()) {
-      EnsureComm(st, GetCtx(entries[0].device), resp.process_set_id);
+      EnsureComm(st, GetCtx(entries[0].device), sid);
     } else {
       ExchangeUniqueId(st, set.ranks.empty() ? 0 : set.ranks[0], "");
     }
-    g_bootstrapped[resp.process_set_id] = true;
+    g_bootstrapped[sid] = true;
+  };
+  if (oneshot_resp && !g_oneshot_bootstrapped.count(sid)) {
+    bool ok = EnsureOneshot(st, entries.empty() ? nullptr
+                                                : &GetCtx(entries[0].device),
+                            sid, !entries.empty());
+    g_oneshot_bootstrapped[sid] = true;
+    // degraded verdict is GLOBAL: every rank (relay included) runs the RCCL
+    // bootstrap now so members can fall back without desyncing the star
+    if (!ok && !g_bootstrapped.count(sid)) rccl_bootstrap();
   }
+  if (!oneshot_resp && !g_bootstrapped.count(sid)) rccl_bootstrap();
   if (entries.empty()) return;  // relay-only rank (not a member)
   if (entries.size() != resp.names.size())
     throw std::runtime_error(
@@ -568,7 +794,14 @@ void Execute(GlobalState& st, Response& resp,
         ") — fusion layout would diverge across ranks");
   int device = entries[0].device;
   auto& ctx = GetCtx(device);
-  ncclComm_t comm = EnsureComm(st, ctx, resp.process_set_id);
+  OneshotCtx* os = nullptr;
+  if (oneshot_resp) {
+    auto oit = ctx.oneshot.find(sid);
+    if (oit != ctx.oneshot.end() && oit->second.ready) os = &oit->second;
+  }
+  // one-shot ops never touch RCCL; every other path (and the degraded
+  // fallback) uses the communicator created by the lock-step bootstrap
+  ncclComm_t comm = os ? nullptr : EnsureComm(st, ctx, sid);
   c10::hip::HIPGuard dguard(device);
   hipStream_t stream = ctx.stream.stream();
   int n = (int)set.ranks.size();
@@ -601,6 +834,38 @@ void Execute(GlobalState& st, Response& resp,
   const char* activity = "RCCL_OP";
   switch (resp.type) {
     case ResponseType::ALLREDUCE: {
+      if (os) {
+        // one-shot over xGMI: pack into my shared slot, peers read all n
+        // slots directly and reduce locally (oneshot.hip protocol)
+        activity = "ONESHOT_ALLREDUCE";
+        os->seq++;
+        int slot = (int)(os->seq % (unsigned long long)os->nslots);
+        OneshotDeviceArgs a{};
+        for (int r = 0; r < os->n; ++r) {
+          a.staging[r] =
+              (char*)os->peer_staging[r] + (int64_t)slot * os->slot_bytes;
+          a.flags[r] = os->peer_flags[r];
+        }
+        a.n = os->n;
+        a.li = os->li;
+        if (os->seq > (unsigned long long)os->nslots)
+          HIP_CHECK(OneshotWaitConsumedLaunch(a, os->seq - os->nslots,
+                                              stream));
+        int64_t total = PackEntries(
+            ctx, entries, wire, false,
+            os->my_staging + (int64_t)slot * os->slot_bytes);
+        auto& fb = FusionBuffer(ctx, total * wire_size);
+        HIP_CHECK(OneshotReduceLaunch(
+            a, os->seq, (int)wire, (unsigned long long)(total * wire_size),
+            fb.data_ptr(), OneshotOpCode(resp.reduce_op), stream));
+        for (auto& e : entries)
+          if (!e.output.defined()) {
+            e.output = at::empty_like(e.tensor);
+            RecordStreamFor(e.output, ctx.stream);
+          }
+        PackEntries(ctx, entries, wire, true);
+        break;
+      }
       activity = "RCCL_ALLREDUCE";
       // dense (possibly permuted, e.g. channels_last) tensors are raw-
       // copyable: identical layout on every rank, so RCCL/pack operate in
@@ -974,12 +1239,18 @@ void Shutdown() {
     for (auto& kv : g_ctx) {
       // comms already aborted (and map cleared) when g_comm_failed
       for (auto& ck : kv.second->comms) ncclCommDestroy(ck.second);
+      for (auto& ok : kv.second->oneshot) {
+        for (void* p : ok.second.opened) (void)hipIpcCloseMemHandle(p);
+        if (ok.second.my_staging) (void)hipFree(ok.second.my_staging);
+        if (ok.second.my_flags) (void)hipFree(ok.second.my_flags);
+      }
       delete kv.second;
     }
     g_ctx.clear();
   }
   g_comm_failed = false;  // elastic re-init starts clean
   g_bootstrapped.clear();
+  g_oneshot_bootstrapped.clear();
   std::lock_guard<std::mutex> g(g_event_mu);
   for (auto ev : g_event_pool) (void)hipEventDestroy(ev);
   g_event_pool.clear();

@@ -40,6 +40,33 @@ hipError_t BatchedCopyLaunch(const CopyBatchArgs& args, int src_dt, int dst_dt,
                              bool with_scale, int blocks_per_copy,
                              hipStream_t stream);
 
+// ---- One-shot xGMI allreduce (oneshot.hip) --------------------------------
+// Every rank reads all n peers' staging slots directly over the
+// fully-connected xGMI mesh and reduces locally — one step instead of a
+// 2(n-1)-hop ring; the win is latency on small/medium buckets.
+constexpr int kOneshotMaxRanks = 8;
+
+struct OneshotDeviceArgs {
+  // set-local-rank-indexed device pointers, slot-adjusted by the host
+  const void* staging[kOneshotMaxRanks];
+  void* flags[kOneshotMaxRanks];  // fine-grained uint64 pages
+  int n = 0;
+  int li = 0;
+};
+
+// Spin (1 tiny block) until every rank's consumed counter reaches min_seq —
+// my staging slot for the next op is free to overwrite.
+hipError_t OneshotWaitConsumedLaunch(const OneshotDeviceArgs& args,
+                                     unsigned long long min_seq,
+                                     hipStream_t stream);
+// Signal ready, wait all ranks, reduce n staging slots into dst (local
+// fusion buffer).  bytes must be 16-aligned (fusion padding guarantees it);
+// op: 0=sum 1=min 2=max 3=product; dt in {DT_F32,DT_F64,DT_F16,DT_BF16}.
+hipError_t OneshotReduceLaunch(const OneshotDeviceArgs& args,
+                               unsigned long long seq, int dt,
+                               unsigned long long bytes, void* dst, int op,
+                               hipStream_t stream);
+
 // ---- Adasum device pipeline (adasum_kernels.hip) ---------------------------
 // Stage 1: per-tensor double-precision dot products / squared norms of (a,b)
 // pairs; Stage 2: a = acoef*a + bcoef*b with coefficients derived on-device.

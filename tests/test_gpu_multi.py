@@ -350,3 +350,25 @@ def test_multi_peer_death_aborts_rccl():
     for r in range(np_ - 1):
         assert "GOT_ERROR" in outs[r], (r, outs[r][-500:])
     assert elapsed < 150, f"survivors took {elapsed:.0f}s to fail"
+
+
+@requires_multi_gpu
+@pytest.mark.parametrize("np_", _nps())
+def test_multi_oneshot_xgmi(np_):
+    """One-shot allreduce over real xGMI peer windows at np>1, mixed with
+    over-threshold buckets that fall back to RCCL on the same stream."""
+    run_workers(np_, CUDA_PRELUDE + """
+        tri = size * (size + 1) // 2
+        for i in range(8):
+            t = torch.full((2048,), float(rank + 1), device=dev)
+            o = hvd.allreduce(t, average=False, name=f"x1.{i}")
+            assert torch.equal(o, torch.full_like(t, float(tri))), i
+        # over threshold (4 MiB default): falls back to RCCL mid-stream
+        big = torch.full((2 << 20,), float(rank + 1), device=dev)
+        ob = hvd.allreduce(big, average=False, name="xbig")
+        assert torch.equal(ob, torch.full_like(big, float(tri)))
+        # back to one-shot after the fallback
+        o2 = hvd.allreduce(torch.full((512,), 1.0, device=dev),
+                           average=False, name="xafter")
+        assert torch.equal(o2, torch.full((512,), float(size), device=dev))
+    """, extra_env={"HOROVOD_ONESHOT_ALLREDUCE": "1"}, timeout=420)
