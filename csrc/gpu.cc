@@ -177,6 +177,61 @@ void ReleaseEvent(hipEvent_t ev) {
 
 void AbortAllCommsLocked(const char* why);
 
+// Poll-based stream sync with a deadline: the RS-VHDD Adasum path blocks
+// the background thread mid-Execute, where the finalizer watchdog cannot
+// see a hung collective — a bounded wait turns a wedged exchange into a
+// failed op instead of a dead process.
+void SyncStreamBounded(hipStream_t stream, double timeout_sec = 300.0) {
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::duration<double>(timeout_sec);
+  hipError_t e;
+  while ((e = hipStreamQuery(stream)) == hipErrorNotReady) {
+    if (g_comm_failed)
+      throw std::runtime_error(
+          "HorovodInternalError: comm aborted during Adasum exchange");
+    if (std::chrono::steady_clock::now() > deadline)
+      throw std::runtime_error("Adasum VHDD exchange timed out");
+    std::this_thread::sleep_for(std::chrono::microseconds(200));
+  }
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("hipStreamQuery: ") +
+                             hipGetErrorString(e));
+}
+
+// Standing async-error watchdog: the finalizer only polls while an op is
+// pending, but a peer can die while the bg thread is blocked inside a
+// synchronous exchange (RS-VHDD) with nothing queued.  This thread checks
+// every live comm's async state at 100 ms cadence.
+struct CommWatchdog {
+  std::thread thread;
+  std::atomic<bool> stop{false};
+  bool started = false;
+} g_watchdog;
+
+void WatchdogLoop() {
+  while (!g_watchdog.stop) {
+    std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    if (g_comm_failed) continue;
+    std::lock_guard<std::mutex> gc(g_ctx_mu);
+    std::lock_guard<std::mutex> g(g_comms_mu);
+    bool bad = false;
+    for (auto& kv : g_ctx) {
+      for (auto& ck : kv.second->comms)
+        if (CommHasAsyncError(ck.second)) bad = true;
+      if (bad) break;
+    }
+    if (bad) AbortAllCommsLocked("RCCL async error (standing watchdog)");
+  }
+}
+
+void EnsureWatchdog() {
+  if (!g_watchdog.started) {
+    g_watchdog.stop = false;
+    g_watchdog.thread = std::thread(WatchdogLoop);
+    g_watchdog.started = true;
+  }
+}
+
 void FinalizerLoop() {
   while (true) {
     PendingOp op;
@@ -216,6 +271,7 @@ void FinalizerLoop() {
       }
       if (g_comm_failed || CommHasAsyncError(op.comm)) {
         {
+          std::lock_guard<std::mutex> gc(g_ctx_mu);
           std::lock_guard<std::mutex> g(g_comms_mu);
           AbortAllCommsLocked("RCCL async error detected by finalizer");
         }
@@ -277,8 +333,9 @@ std::string ExchangeUniqueId(GlobalState& st, int leader,
   return st.comm.Bcast(idb);
 }
 
-// Abort every live communicator (called with g_comms_mu held).  After this,
-// all GPU collectives fail fast until the next elastic re-init.
+// Abort every live communicator (called with g_ctx_mu AND g_comms_mu held,
+// in that order).  After this, all GPU collectives fail fast until the
+// next elastic re-init.
 void AbortAllCommsLocked(const char* why) {
   if (g_comm_failed.exchange(true)) return;  // once
   HVD_LOG(ERROR, "aborting all RCCL comms: %s", why);
@@ -329,6 +386,7 @@ ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
     ctx.comms[set_id] = comm;
   }
   g_rccl_used = true;
+  EnsureWatchdog();
   HVD_LOG(INFO, "RCCL comm ready: process set %d, %d ranks, device %d",
           (int)set_id, (int)set.ranks.size(), ctx.device);
   return comm;
@@ -651,6 +709,221 @@ int64_t PackEntries(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
 }
 
 
+// ---- Distributed VHDD Adasum (reference: adasum_gpu_operations.cc:44-120
+// structure + adasum.h:195-345 FusedAllreduce) -----------------------------
+//
+// Vector-halving distance-doubling on the fused buffer: at level k each
+// rank exchanges half of its current piece with partner li^2^k over RCCL
+// point-to-point, computes per-tensor partial dot/norm products on its
+// kept half (CDNA4 kernels), sums the scalars across the sharing group
+// over the control star (3 doubles/tensor — tiny), and applies the
+// projection-weighted combine in place.  O(total) memory and O(total)
+// compute per rank vs the one-shot tree's n x both; the one-shot tree
+// (below) remains the small-bucket option.
+//
+// Only valid for the GLOBAL process set: every global rank must relay the
+// per-level scalar star rounds, and non-members of a subset op return
+// before Execute's op switch.
+
+int64_t AdasumOneshotThreshold() {
+  static int64_t v = [] {
+    const char* e = std::getenv("HOROVOD_ADASUM_ONESHOT_THRESHOLD");
+    return e ? (int64_t)atoll(e) : (int64_t)(1 << 20);
+  }();
+  return v;
+}
+
+void ExecuteAdasumRSVHDD(GlobalState& st, DeviceCtx& ctx, Response& resp,
+                         std::vector<TensorTableEntry>& entries,
+                         ncclComm_t comm) {
+  auto& set = st.controller->process_set(resp.process_set_id);
+  const int n = (int)set.ranks.size();
+  const int li = set.local_index(st.rank);
+  DataType wire = resp.dtype;
+  const int64_t wsz = (int64_t)DataTypeSize(wire);
+  hipStream_t stream = ctx.stream.stream();
+  auto wire_nccl = ToNccl(wire);
+
+  const int64_t L = PackEntries(ctx, entries, wire, false);  // 64-aligned
+  char* base = (char*)ctx.fusion_buffer.data_ptr();
+
+  int p = 1;
+  while (p * 2 <= n) p *= 2;
+  int levels = 0;
+  while ((1 << levels) < p) ++levels;
+
+  // scratch for received pieces (<= L elems), device dots (3/tensor)
+  if (!ctx.adasum_buffer.defined() || ctx.adasum_buffer.numel() < L * wsz) {
+    c10::hip::HIPStreamGuard sg(ctx.stream);
+    ctx.adasum_buffer = at::empty(
+        {L * wsz},
+        at::TensorOptions().dtype(at::kByte).device(at::kCUDA, ctx.device));
+  }
+  char* scratch = (char*)ctx.adasum_buffer.data_ptr();
+  const int64_t T = (int64_t)entries.size();
+  const int64_t ndots = T * 3;
+  if (!ctx.dots_buffer.defined() || ctx.dots_buffer.numel() < ndots) {
+    c10::hip::HIPStreamGuard sg(ctx.stream);
+    ctx.dots_buffer = at::empty(
+        {ndots},
+        at::TensorOptions().dtype(at::kDouble).device(at::kCUDA, ctx.device));
+  }
+  double* ddots = ctx.dots_buffer.data_ptr<double>();
+
+  // per-tensor REAL ranges in fused-element coordinates (padding excluded
+  // from dot products — reference per-tensor coefficient isolation)
+  std::vector<int64_t> offs(T), cnts(T);
+  {
+    int64_t off = 0;
+    for (int64_t t = 0; t < T; ++t) {
+      offs[t] = off;
+      cnts[t] = entries[t].tensor.numel();
+      off += AlignedElems(cnts[t]);
+    }
+  }
+
+  // run the per-tensor kernels over the intersection of piece [S,E) with
+  // each tensor's real range; a_is_mine controls canonical (a=lower-rank)
+  // pointer order for dots, and which buffer ScaledAdd mutates
+  auto run_frags = [&](int64_t S, int64_t E, bool mine_is_a, bool scaled_add,
+                       int64_t piece_base /* scratch index of S */) {
+    AdasumBatchArgs args;
+    args.count = 0;
+    int64_t dot_idx0 = -1;
+    auto flush = [&](bool add) {
+      if (args.count == 0) return;
+      if (add)
+        HIP_CHECK(AdasumScaledAddLaunch(args, (int)wire,
+                                        ddots + dot_idx0 * 3, stream));
+      else
+        HIP_CHECK(AdasumDotsLaunch(args, (int)wire, ddots + dot_idx0 * 3,
+                                   stream));
+      args.count = 0;
+      dot_idx0 = -1;
+    };
+    for (int64_t t = 0; t < T; ++t) {
+      int64_t lo = std::max(S, offs[t]);
+      int64_t hi = std::min(E, offs[t] + cnts[t]);
+      char* mine = base + lo * wsz;
+      char* theirs = scratch + (lo - S + piece_base) * wsz;
+      // every tensor keeps a dots slot (group sums stay aligned), even when
+      // this piece doesn't intersect it (numel 0)
+      if (args.count == kCopyBatchCapacity) flush(scaled_add);
+      if (dot_idx0 < 0) dot_idx0 = t;
+      int k = args.count++;
+      if (scaled_add) {
+        // result goes into MY buffer regardless of role; the host swapped
+        // na/nb for the higher rank so the kernel's coefficient order is
+        // right for (a=mine, b=theirs)
+        args.a[k] = mine;
+        args.b[k] = theirs;
+      } else {
+        args.a[k] = mine_is_a ? (void*)mine : (void*)theirs;
+        args.b[k] = mine_is_a ? (const void*)theirs : (const void*)mine;
+      }
+      args.numel[k] = (unsigned long long)std::max<int64_t>(0, hi - lo);
+    }
+    flush(scaled_add);
+  };
+
+  // star exchange of per-tensor partial dots; every global rank calls this
+  // exactly `levels` times (active ranks with data, folded ranks empty)
+  auto scalar_round = [&](std::vector<double>& partials, int group_lo,
+                          int group_hi) {
+    std::string payload;
+    if (!partials.empty())
+      payload.assign((const char*)partials.data(),
+                     partials.size() * sizeof(double));
+    auto all = ExchangeAllGather(st, payload);
+    if (partials.empty()) return;
+    std::fill(partials.begin(), partials.end(), 0.0);
+    for (int r = group_lo; r < group_hi; ++r) {
+      int grank = set.ranks[r];
+      if (grank >= (int)all.size()) continue;
+      auto& s = all[grank];
+      if (s.size() != partials.size() * sizeof(double)) continue;
+      const double* d = (const double*)s.data();
+      for (size_t i = 0; i < partials.size(); ++i) partials[i] += d[i];
+    }
+  };
+
+  // ---- fold the non-power-of-2 remainder: pair (i-p, i), local dots -----
+  if (li >= p) {
+    RCCL_CHECK(ncclGroupStart());
+    RCCL_CHECK(ncclSend(base, L, wire_nccl, li - p, comm, stream));
+    RCCL_CHECK(ncclGroupEnd());
+  } else if (li + p < n) {
+    RCCL_CHECK(ncclGroupStart());
+    RCCL_CHECK(ncclRecv(scratch, L, wire_nccl, li + p, comm, stream));
+    RCCL_CHECK(ncclGroupEnd());
+    HIP_CHECK(hipMemsetAsync(ddots, 0, (size_t)ndots * sizeof(double),
+                             stream));
+    run_frags(0, L, /*mine_is_a=*/true, /*scaled_add=*/false, 0);
+    run_frags(0, L, true, /*scaled_add=*/true, 0);
+  }
+
+  // ---- VHDD levels -------------------------------------------------------
+  int64_t S = 0, len = L;
+  std::vector<double> host_dots((size_t)ndots);
+  for (int k = 0; k < levels; ++k) {
+    int stride = 1 << k;
+    if (li < p) {
+      int partner = li ^ stride;
+      bool lower = li < partner;
+      int64_t half = len / 2;
+      int64_t myS = lower ? S : S + half;
+      int64_t thS = lower ? S + half : S;
+      RCCL_CHECK(ncclGroupStart());
+      RCCL_CHECK(ncclSend(base + thS * wsz, half, wire_nccl, partner, comm,
+                          stream));
+      RCCL_CHECK(ncclRecv(scratch, half, wire_nccl, partner, comm, stream));
+      RCCL_CHECK(ncclGroupEnd());
+      HIP_CHECK(hipMemsetAsync(ddots, 0, (size_t)ndots * sizeof(double),
+                               stream));
+      run_frags(myS, myS + half, lower, false, 0);
+      HIP_CHECK(hipMemcpyAsync(host_dots.data(), ddots,
+                               (size_t)ndots * sizeof(double),
+                               hipMemcpyDeviceToHost, stream));
+      SyncStreamBounded(stream);
+      int group = 2 * stride;
+      int g0 = (li / group) * group;
+      scalar_round(host_dots, g0, g0 + group);
+      if (!lower)
+        for (int64_t t = 0; t < T; ++t)
+          std::swap(host_dots[t * 3 + 1], host_dots[t * 3 + 2]);
+      HIP_CHECK(hipMemcpyAsync(ddots, host_dots.data(),
+                               (size_t)ndots * sizeof(double),
+                               hipMemcpyHostToDevice, stream));
+      run_frags(myS, myS + half, lower, true, 0);
+      S = myS;
+      len = half;
+    } else {
+      // folded rank: relay the scalar star round in lock-step
+      std::vector<double> none;
+      scalar_round(none, 0, 0);
+    }
+  }
+
+  // ---- regather: grouped broadcast of the p final pieces -----------------
+  const int64_t piece = L / p;
+  RCCL_CHECK(ncclGroupStart());
+  for (int r = 0; r < p; ++r) {
+    int64_t Sr = 0;
+    for (int k = 0; k < levels; ++k)
+      if ((r >> k) & 1) Sr += L >> (k + 1);
+    RCCL_CHECK(ncclBroadcast(base + Sr * wsz, base + Sr * wsz, piece,
+                             wire_nccl, r, comm, stream));
+  }
+  RCCL_CHECK(ncclGroupEnd());
+
+  for (auto& e : entries)
+    if (!e.output.defined()) {
+      e.output = at::empty_like(e.tensor);
+      RecordStreamFor(e.output, ctx.stream);
+    }
+  PackEntries(ctx, entries, wire, true);
+}
+
 // One-shot Adasum over xGMI: allgather every rank's fused buffer (7 wide
 // point-to-point links make this efficient on one MI355X node), then run the
 // VHDD combine tree LOCALLY with the CDNA4 dot/scaled-add kernels — a
@@ -661,6 +934,24 @@ void ExecuteAdasum(GlobalState& st, DeviceCtx& ctx, Response& resp,
                    std::vector<TensorTableEntry>& entries, ncclComm_t comm) {
   auto& set = st.controller->process_set(resp.process_set_id);
   int n = (int)set.ranks.size();
+  if (n > 1 && set.ranks.size() == (size_t)st.size &&
+      std::getenv("HOROVOD_ADASUM_ONESHOT") == nullptr) {
+    // rank-consistent size check from the response (every member agrees)
+    int64_t aligned = 0;
+    const int64_t* ps = resp.tensor_shapes.data();
+    const int64_t* pe = ps + resp.tensor_shapes.size();
+    while (ps < pe) {
+      int64_t nd = *ps++;
+      int64_t numel = 1;
+      for (int64_t i = 0; i < nd; ++i) numel *= *ps++;
+      aligned += AlignedElems(numel);
+    }
+    if (aligned * (int64_t)DataTypeSize(resp.dtype) >=
+        AdasumOneshotThreshold()) {
+      ExecuteAdasumRSVHDD(st, ctx, resp, entries, comm);
+      return;
+    }
+  }
   DataType wire = resp.dtype;
   int64_t wire_size = (int64_t)DataTypeSize(wire);
   hipStream_t stream = ctx.stream.stream();
@@ -1067,6 +1358,7 @@ void Execute(GlobalState& st, Response& resp,
 }
 
 void AbortComms(const std::string& why) {
+  std::lock_guard<std::mutex> gc(g_ctx_mu);
   std::lock_guard<std::mutex> g(g_comms_mu);
   AbortAllCommsLocked(why.c_str());
 }
@@ -1225,6 +1517,11 @@ void WaitAllPending() {
 
 void Shutdown() {
   WaitAllPending();
+  if (g_watchdog.started) {
+    g_watchdog.stop = true;
+    if (g_watchdog.thread.joinable()) g_watchdog.thread.join();
+    g_watchdog.started = false;
+  }
   {
     std::lock_guard<std::mutex> g(g_finalizer.mu);
     g_finalizer.stop = true;

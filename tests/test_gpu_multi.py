@@ -372,3 +372,43 @@ def test_multi_oneshot_xgmi(np_):
                            average=False, name="xafter")
         assert torch.equal(o2, torch.full((512,), float(size), device=dev))
     """, extra_env={"HOROVOD_ONESHOT_ALLREDUCE": "1"}, timeout=420)
+
+
+@requires_multi_gpu
+@pytest.mark.parametrize("np_", _nps())
+def test_multi_adasum_rsvhdd(np_):
+    """Large-bucket Adasum takes the distributed VHDD path (RCCL halving
+    exchange + star scalar sums, O(total) memory) — same float64 golden
+    tree as the one-shot test, above the 1 MiB threshold."""
+    run_workers(np_, CUDA_PRELUDE + """
+        n = size
+        N = 400_000  # 1.6 MB fp32 > HOROVOD_ADASUM_ONESHOT_THRESHOLD
+        def vec(r):
+            i = torch.arange(N, dtype=torch.float64)
+            return torch.sin(i * 0.001 * (r + 1)) + 0.01 * (r + 1)
+        def combine(a, b):
+            dot = (a * b).sum(); na = (a * a).sum(); nb = (b * b).sum()
+            ac = 1.0 - dot / (2 * na) if na > 0 else 1.0
+            bc = 1.0 - dot / (2 * nb) if nb > 0 else 1.0
+            return a * ac + b * bc
+        work = [vec(r) for r in range(n)]
+        p = 1
+        while p * 2 <= n:
+            p *= 2
+        for i in range(p, n):
+            work[i - p] = combine(work[i - p], work[i])
+        stride = 1
+        while stride < p:
+            for i in range(0, p - stride, 2 * stride):
+                work[i] = combine(work[i], work[i + stride])
+            stride *= 2
+        expected = work[0].float()
+        out = hvd.allreduce(vec(rank).float().to(dev), op=hvd.Adasum,
+                            name="rsv")
+        err = (out.cpu().double() - work[0]).abs().max().item()
+        assert err < 1e-4, err
+        # repeat to exercise buffer reuse
+        out2 = hvd.allreduce(vec(rank).float().to(dev), op=hvd.Adasum,
+                             name="rsv2")
+        assert torch.equal(out, out2)
+    """, timeout=420)
