@@ -206,8 +206,11 @@ Controller::Controller(StarComm* comm, int rank, int size, ControllerConfig cfg)
   start_time_ = std::chrono::steady_clock::now();
   if (rank_ == 0 && size_ > 1 && std::getenv("HOROVOD_AUTOTUNE")) {
     const char* lp = std::getenv("HOROVOD_AUTOTUNE_LOG");
+    const char* ot = std::getenv("HOROVOD_ONESHOT_THRESHOLD");
     autotuner_.reset(new Autotuner(cfg_.fusion_threshold_bytes,
-                                   cfg_.cycle_time_ms, lp ? lp : ""));
+                                   cfg_.cycle_time_ms,
+                                   ot ? atoll(ot) : (int64_t)(4 << 20),
+                                   lp ? lp : ""));
   }
   ProcessSetInfo global;
   global.id = 0;
@@ -247,7 +250,8 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
       slow.push_back(std::move(req));
       continue;
     }
-    auto state = cache_.Lookup(req);
+    auto state = cache_enabled_ ? cache_.Lookup(req)
+                                : ResponseCache::State::MISS;
     if (state == ResponseCache::State::HIT) {
       cached_pending_.push_back(std::move(req));
     } else {
@@ -402,7 +406,7 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
         if (resp.type == ResponseType::JOIN || resp.type == ResponseType::BARRIER)
           inflight_.erase(SetKey(resp.process_set_id,
                                  resp.type == ResponseType::JOIN ? "join" : "barrier"));
-        {
+        if (cache_enabled_) {
           int32_t li = -1;
           auto its = process_sets_.find(resp.process_set_id);
           if (its != process_sets_.end()) li = its->second.local_index(rank_);
@@ -438,7 +442,7 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
   // ---- autotuner (rank 0): score this cycle's reduced bytes; a ready
   // proposal is published next cycle as a TUNE response (slow path forced
   // via pending_tune_).
-  if (autotuner_ && !autotuner_->done()) {
+  if (autotuner_) {
     int64_t bytes = 0;
     for (auto& resp : result.responses) {
       if (resp.type != ResponseType::ALLREDUCE &&
@@ -456,7 +460,9 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     double now = std::chrono::duration<double>(std::chrono::steady_clock::now() -
                                                start_time_)
                      .count();
-    if (autotuner_->Record(bytes, now)) pending_tune_ = true;
+    bool propose = autotuner_->done() ? autotuner_->Watch(bytes, now)
+                                      : autotuner_->Record(bytes, now);
+    if (propose) pending_tune_ = true;
   }
   return result;
 }
@@ -571,7 +577,8 @@ std::vector<Response> Controller::CoordinatorProcess(
     Response t;
     t.type = ResponseType::TUNE;
     auto p = autotuner_->current();
-    t.tensor_sizes = {p.fusion_bytes, (int64_t)(p.cycle_time_ms * 1000.0)};
+    t.tensor_sizes = {p.fusion_bytes, (int64_t)(p.cycle_time_ms * 1000.0),
+                      p.oneshot_threshold, p.cache_enabled ? 1 : 0};
     out.push_back(std::move(t));
     pending_tune_ = false;
   }

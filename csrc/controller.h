@@ -140,9 +140,11 @@ class Controller {
   // tensors still negotiating (cache-hit bits not yet agreed, or requests at
   // the coordinator) — the cycle loop must not sleep while any exist
   bool has_pending() const { return !cached_pending_.empty() || !inflight_.empty(); }
-  void ApplyTune(int64_t fusion_bytes, double cycle_time_ms) {
+  void ApplyTune(int64_t fusion_bytes, double cycle_time_ms,
+                 bool cache_enabled = true) {
     cfg_.fusion_threshold_bytes = fusion_bytes;
     cfg_.cycle_time_ms = cycle_time_ms;
+    cache_enabled_ = cache_enabled;
   }
 
  private:
@@ -180,6 +182,9 @@ class Controller {
   // autotuner (rank 0 only; results propagate via TUNE responses)
   std::unique_ptr<Autotuner> autotuner_;
   bool pending_tune_ = false;
+  // autotuner categorical arm: when off, every request renegotiates (the
+  // reference's cache on/off boolean); toggled rank-synchronously via TUNE
+  bool cache_enabled_ = true;
 
  public:
   // join bookkeeping shared with core

@@ -795,9 +795,13 @@ void PerformOperation(GlobalState& st, Response& resp) {
   }
 
   if (resp.type == ResponseType::TUNE) {
-    if (resp.tensor_sizes.size() == 2)
+    if (resp.tensor_sizes.size() >= 2) {
+      bool cache_on = resp.tensor_sizes.size() < 4 || resp.tensor_sizes[3];
       st.controller->ApplyTune(resp.tensor_sizes[0],
-                               resp.tensor_sizes[1] / 1000.0);
+                               resp.tensor_sizes[1] / 1000.0, cache_on);
+      if (resp.tensor_sizes.size() >= 3)
+        gpu::SetOneshotThreshold(resp.tensor_sizes[2]);
+    }
     return;
   }
   if (resp.type == ResponseType::JOIN) {

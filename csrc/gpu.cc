@@ -400,12 +400,28 @@ std::unordered_map<int32_t, bool> g_oneshot_bootstrapped;
 
 // ---- one-shot allreduce host side -----------------------------------------
 
-int64_t OneshotThreshold() {
+// Crossover threshold is autotuner-adjustable at runtime (TUNE responses
+// apply it rank-synchronously in response order, so eligibility stays
+// consistent across ranks).  Staging capacity is fixed at the MAX so a
+// raised threshold never overflows the IPC windows.
+std::atomic<int64_t> g_oneshot_threshold{-1};
+
+int64_t OneshotThresholdMax() {
   static int64_t v = [] {
-    const char* e = std::getenv("HOROVOD_ONESHOT_THRESHOLD");
-    return e ? (int64_t)atoll(e) : (int64_t)(4 << 20);
+    const char* e = std::getenv("HOROVOD_ONESHOT_THRESHOLD_MAX");
+    return e ? (int64_t)atoll(e) : (int64_t)(8 << 20);
   }();
   return v;
+}
+
+int64_t OneshotThreshold() {
+  int64_t v = g_oneshot_threshold.load(std::memory_order_relaxed);
+  if (v < 0) {
+    const char* e = std::getenv("HOROVOD_ONESHOT_THRESHOLD");
+    v = e ? (int64_t)atoll(e) : (int64_t)(4 << 20);
+    g_oneshot_threshold.store(v, std::memory_order_relaxed);
+  }
+  return std::min(v, OneshotThresholdMax());
 }
 
 bool OneshotEnabled() {
@@ -494,7 +510,7 @@ bool EnsureOneshot(GlobalState& st, DeviceCtx* ctx, int32_t set_id,
     auto& set = st.controller->process_set(set_id);
     os->n = (int)set.ranks.size();
     os->li = set.local_index(st.rank);
-    os->slot_bytes = OneshotThreshold();
+    os->slot_bytes = OneshotThresholdMax();
     c10::hip::HIPGuard guard(ctx->device);
     hipIpcMemHandle_t hs{}, hf{};
     do {
@@ -1364,6 +1380,10 @@ void AbortComms(const std::string& why) {
 }
 
 bool CommsFailed() { return g_comm_failed; }
+
+void SetOneshotThreshold(int64_t bytes) {
+  g_oneshot_threshold.store(bytes, std::memory_order_relaxed);
+}
 
 void AdasumCombine(std::vector<at::Tensor>& a, std::vector<at::Tensor>& b) {
   if (a.empty()) return;

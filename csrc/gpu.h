@@ -41,6 +41,10 @@ void AbortComms(const std::string& why);
 // True once AbortComms ran (cleared by Shutdown for elastic re-init).
 bool CommsFailed();
 
+// Autotuner hook: adjust the ring-vs-one-shot crossover at runtime
+// (applied rank-synchronously from TUNE responses).
+void SetOneshotThreshold(int64_t bytes);
+
 // True once a RCCL communicator has been created (used by tests to assert
 // the native path ran).
 bool RcclUsed();

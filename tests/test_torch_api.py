@@ -572,3 +572,30 @@ def test_groups_with_backward_passes_np2():
         assert torch.allclose(g[0], g[1], atol=1e-6), \
             (g[0] - g[1]).abs().max()
     """)
+
+
+def test_autotune_categorical_np2(tmp_path):
+    """Round-2: the autotuner explores the categorical arms (cache on/off +
+    one-shot crossover) — log rows carry 5 fields and results stay correct
+    while the cache-off arm renegotiates every cycle."""
+    log = str(tmp_path / "atc.csv")
+    run_workers(2, """
+        import time
+        deadline = time.time() + 25
+        i = 0
+        while time.time() < deadline and i < 6000:
+            out = hvd.allreduce(torch.full((2048,), float(rank + 1)),
+                                average=False, name="atc")
+            assert out[0].item() == 3.0, out[0].item()
+            i += 1
+    """, extra_env={"HOROVOD_AUTOTUNE": "1",
+                    "HOROVOD_AUTOTUNE_WINDOW_SECONDS": "0.1",
+                    "HOROVOD_AUTOTUNE_LOG": log}, timeout=300)
+    rows = [ln for ln in open(log).read().splitlines()
+            if ln and not ln.startswith(("CONVERGED", "REOPENED"))]
+    assert rows, "autotune log empty"
+    assert all(len(r.split(",")) == 5 for r in rows), rows[:3]
+    # the warm-start grid pins a cache-off probe; with >= 13 windows it ran
+    if len(rows) >= 13:
+        assert any(r.split(",")[3] == "0" for r in rows), \
+            "cache-off arm never explored"
