@@ -64,6 +64,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stall_warning_sec"), py::arg("stall_shutdown_sec"),
         py::arg("timeline"));
 
+  m.def("interrupt", [](const std::string& why) {
+    py::gil_scoped_release nogil;
+    InterruptHorovod(why);
+  });
   m.def("shutdown", [] {
     py::gil_scoped_release nogil;
     ShutdownHorovod();

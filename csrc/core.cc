@@ -1047,6 +1047,15 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
           rank, size, local_rank, local_size, addr.c_str(), port);
 }
 
+void InterruptHorovod(const std::string& why) {
+  auto& st = State();
+  if (!st.initialized || st.aborted) return;
+  // prompt elastic scale-down: a displaced worker (or a host-side watchdog)
+  // fails all in-flight and future collectives with HorovodInternalError so
+  // the retry loop resets without waiting for the next commit()
+  Abort(st, why);
+}
+
 void ShutdownHorovod() {
   auto& st = State();
   if (!st.initialized) return;

@@ -123,6 +123,8 @@ void InitHorovod(int rank, int size, int local_rank, int local_size,
                  int cross_rank, int cross_size, const std::string& addr,
                  int port, const ControllerConfig& cfg);
 void ShutdownHorovod();
+// Fail all in-flight + future collectives with ABORTED (elastic interrupt).
+void InterruptHorovod(const std::string& why);
 bool IsInitialized();
 // Cut the current cycle-pacing wait (the caller is about to block on
 // results): lets the background thread negotiate the tail immediately.

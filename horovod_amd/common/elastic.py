@@ -45,9 +45,29 @@ class _WorkerNotificationManager:
                 v = rendezvous_get("version").get("version", self._known_version)
                 if v > self._known_version:
                     self._updated.set()
+                    self._interrupt_if_displaced()
             except Exception:
                 pass
-            time.sleep(1.0)
+            time.sleep(0.5)
+
+    def _interrupt_if_displaced(self):
+        """Prompt scale-down (round-2): a worker dropped from the new
+        assignment must stop within ~1 s regardless of commit cadence
+        (reference pushes notifications, worker.py:46-119; round-1 only
+        noticed at the next commit()).  Interrupting the native core fails
+        its next collective with HorovodInternalError, so the retry loop
+        resets immediately and the rendezvous answers `terminate`."""
+        try:
+            wid = os.environ.get("HOROVOD_WORKER_ID")
+            if not wid:
+                return
+            a = rendezvous_get("assignment", worker_id=wid,
+                               after=self._known_version)
+            if a.get("terminate"):
+                from horovod_amd import _core
+                _core.interrupt("displaced by elastic host update")
+        except Exception:
+            pass
 
     def host_updates_available(self):
         return self._updated.is_set()

@@ -28,6 +28,7 @@ opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters())
 state = elastic.TorchState(model, opt, batch=0)
 
 TARGET = int(os.environ.get("TEST_TARGET_BATCHES", "30"))
+COMMIT_EVERY = int(os.environ.get("TEST_COMMIT_EVERY", "1"))
 CRASH_AT = int(os.environ.get("TEST_CRASH_AT", "-1"))
 CRASH_WID = os.environ.get("TEST_CRASH_WID", "")
 MARKER = os.environ.get("TEST_MARKER_FILE", "")
@@ -46,7 +47,8 @@ def train(state):
         loss.backward()
         opt.step()
         state.batch += 1
-        state.commit()
+        if state.batch % COMMIT_EVERY == 0:
+            state.commit()
 
 train(state)
 with open(MARKER, "a") as f:
@@ -106,6 +108,29 @@ def test_elastic_scale_up(tmp_path):
     assert err is None, err
     text = open(marker).read()
     assert "size=2" in text, text
+
+
+def test_elastic_prompt_scale_down(tmp_path):
+    """Round-2: a worker dropped from the assignment stops within seconds
+    even when it NEVER commits (the notification poll interrupts the native
+    core; round-1 only noticed displacement at the next commit())."""
+    driver, disc, marker = _driver(
+        tmp_path, {"127.0.0.1": 2},
+        extra_env={"TEST_TARGET_BATCHES": "5000",
+                   "TEST_COMMIT_EVERY": "100000"})
+    driver.start()
+    time.sleep(3.0)
+    displaced = driver._workers.get("127.0.0.1:1")
+    assert displaced is not None and displaced.poll() is None
+    disc.set({"127.0.0.1": 1})
+    t0 = time.time()
+    while displaced.poll() is None and time.time() - t0 < 30:
+        time.sleep(0.25)
+    elapsed = time.time() - t0
+    driver.stop()
+    assert displaced.poll() is not None, \
+        f"displaced worker still running {elapsed:.0f}s after scale-down"
+    assert elapsed < 30, elapsed
 
 
 def test_elastic_min_np_violation(tmp_path):
