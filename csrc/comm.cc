@@ -174,14 +174,24 @@ void StarComm::Barrier() {
   Bcast("");
 }
 
-void StarComm::BitAnd(std::string& bits) {
+void StarComm::BitAnd(std::string& bits, size_t verify_tail) {
   auto all = Gather(bits);
   if (rank_ == 0) {
     std::string acc = all[0];
+    size_t head = acc.size() - std::min(verify_tail, acc.size());
     for (int r = 1; r < size_; ++r) {
       if (all[r].size() != acc.size())
         throw std::runtime_error("horovod_amd comm: bitvector length mismatch");
-      for (size_t i = 0; i < acc.size(); ++i) acc[i] &= all[r][i];
+      for (size_t i = 0; i < head; ++i) acc[i] &= all[r][i];
+      for (size_t i = head; i < acc.size(); ++i) {
+        if (all[r][i] != acc[i])
+          throw std::runtime_error(
+              "horovod_amd: response-stream divergence detected — rank " +
+              std::to_string(r) + "'s executed-response hash differs from "
+              "rank 0's (fast-path cache state desynchronized; this would "
+              "deadlock inside RCCL). Set HOROVOD_CACHE_CAPACITY=0 or "
+              "report a bug.");
+      }
     }
     bits = Bcast(acc);
   } else {

@@ -38,8 +38,12 @@ class StarComm {
   std::string Bcast(const std::string& payload);
   void Barrier();
   // Byte-wise AND / OR across ranks (cache-bitvector sync; reference:
-  // mpi_controller.cc CrossRankBitwiseAnd/Or).
-  void BitAnd(std::string& bits);
+  // mpi_controller.cc CrossRankBitwiseAnd/Or).  When verify_tail > 0 the
+  // last verify_tail bytes of the frame are NOT reduced: the root asserts
+  // they are identical on every rank and throws on divergence (the
+  // response-stream hash cross-check — one divergent fast-path cache state
+  // would otherwise deadlock inside RCCL with no diagnosis).
+  void BitAnd(std::string& bits, size_t verify_tail = 0);
   void BitOr(std::string& bits);
   // Root sends frames[r] to each rank r; every rank returns its own frame.
   std::string ScatterFrames(const std::vector<std::string>& frames);

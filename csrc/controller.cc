@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstring>
 #include <sstream>
 
 #include "core.h"
@@ -203,6 +204,7 @@ void ResponseCache::Put(const Response& response, const std::vector<Request>& re
 Controller::Controller(StarComm* comm, int rank, int size, ControllerConfig cfg)
     : comm_(comm), rank_(rank), size_(size), cfg_(cfg) {
   cache_.set_capacity(cfg_.cache_capacity);
+  check_stream_ = std::getenv("HOROVOD_CHECK_RESPONSE_STREAM") != nullptr;
   start_time_ = std::chrono::steady_clock::now();
   if (rank_ == 0 && size_ > 1 && std::getenv("HOROVOD_AUTOTUNE")) {
     const char* lp = std::getenv("HOROVOD_AUTOTUNE_LOG");
@@ -280,7 +282,8 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
   ResponseList result;
 
   if (size_ > 1) {
-    std::string vecA(nbytes + 1, '\0');
+    const size_t tail = check_stream_ ? sizeof(uint64_t) : 0;
+    std::string vecA(nbytes + 1 + tail, '\0');
     // A rank votes "ready" for a slot if it has the tensor queued OR it is
     // not a member of the slot's process set (so subset tensors can take the
     // fast path under a global AND).
@@ -300,7 +303,9 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
       if (vote) vecA[s / 8] |= (char)(1 << (s % 8));
     }
     if (shutdown_requested) vecA[nbytes] |= 1;
-    comm_->BitAnd(vecA);
+    if (tail)
+      std::memcpy(&vecA[nbytes + 1], &response_hash_, sizeof(uint64_t));
+    comm_->BitAnd(vecA, tail);
 
     std::string vecB(nbytes + 1, '\0');
     for (int slot : my_invalid_slots) vecB[slot / 8] |= (char)(1 << (slot % 8));
@@ -437,6 +442,18 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     cached_pending_.clear();
     auto fused = FuseResponses(ready);
     for (auto& r : fused) result.responses.push_back(std::move(r));
+  }
+
+  if (check_stream_) {
+    // FNV-1a over each response's wire serialization, in execution order
+    for (auto& resp : result.responses) {
+      std::string ser;
+      resp.Serialize(ser);
+      for (unsigned char c : ser) {
+        response_hash_ ^= c;
+        response_hash_ *= 1099511628211ull;
+      }
+    }
   }
 
   // ---- autotuner (rank 0): score this cycle's reduced bytes; a ready

@@ -185,6 +185,11 @@ class Controller {
   // autotuner categorical arm: when off, every request renegotiates (the
   // reference's cache on/off boolean); toggled rank-synchronously via TUNE
   bool cache_enabled_ = true;
+  // HOROVOD_CHECK_RESPONSE_STREAM=1: running FNV-1a hash of every executed
+  // response, cross-checked in the bitvector round each cycle — catches a
+  // diverging fast path before it deadlocks RCCL
+  bool check_stream_ = false;
+  uint64_t response_hash_ = 1469598103934665603ull;
 
  public:
   // join bookkeeping shared with core
