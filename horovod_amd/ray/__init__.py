@@ -81,3 +81,97 @@ class RayExecutor:
         for w in self.workers:
             ray.kill(w)
         self.workers = []
+
+
+class RayHostDiscovery:
+    """HostDiscovery over Ray's cluster view (reference:
+    ray/elastic_v2.py:40-110 RayHostDiscovery): alive nodes -> slots from
+    their GPU (or CPU) resources."""
+
+    def __init__(self, use_gpu=False, cpus_per_worker=1, gpus_per_worker=1):
+        self.use_gpu = use_gpu
+        self.cpus_per_worker = cpus_per_worker
+        self.gpus_per_worker = gpus_per_worker
+
+    def find_available_hosts_and_slots(self):
+        ray = _require_ray()
+        hosts = {}
+        for node in ray.nodes():
+            if not node.get("Alive"):
+                continue
+            res = node.get("Resources", {})
+            host = node.get("NodeManagerAddress")
+            if self.use_gpu:
+                slots = int(res.get("GPU", 0) // max(self.gpus_per_worker, 1))
+            else:
+                slots = int(res.get("CPU", 0) // max(self.cpus_per_worker, 1))
+            if host and slots > 0:
+                hosts[host] = slots
+        return hosts
+
+
+class ElasticRayExecutor:
+    """Elastic training on a Ray cluster (reference: ray/elastic_v2.py
+    ElasticAdapter): the elastic driver follows Ray's node membership via
+    RayHostDiscovery; workers are launched onto the discovered hosts with
+    the standard elastic rendezvous env (ssh/local spawn — every Ray node
+    of a typical cluster is ssh-reachable, and single-node clusters spawn
+    locally)."""
+
+    def __init__(self, use_gpu=False, cpus_per_worker=1, gpus_per_worker=1,
+                 min_np=1, max_np=None, reset_limit=None,
+                 override_discovery=None):
+        self.discovery = override_discovery or RayHostDiscovery(
+            use_gpu, cpus_per_worker, gpus_per_worker)
+        self.min_np = min_np
+        self.max_np = max_np
+        self.reset_limit = reset_limit
+        self._driver = None
+
+    def start(self):
+        # fail early with a clear error if ray is absent (unless a custom
+        # discovery was injected, e.g. in tests or hybrid clusters)
+        if isinstance(self.discovery, RayHostDiscovery):
+            _require_ray()
+
+    def run(self, fn, args=(), kwargs=None):
+        """Run `fn` (an hvd.elastic-decorated callable or any function using
+        the elastic API) across the cluster until completion."""
+        import pickle
+        import sys
+        import tempfile
+
+        import cloudpickle
+
+        from horovod_amd.runner.elastic_driver import ElasticDriver
+
+        payload = cloudpickle.dumps((fn, tuple(args), dict(kwargs or {})))
+        fd, fn_path = tempfile.mkstemp(suffix=".pkl", dir=os.getcwd())
+        with os.fdopen(fd, "wb") as f:
+            f.write(payload)
+        out_path = fn_path + ".out"
+        command = [sys.executable, "-m", "horovod_amd.spark._elastic_task",
+                   fn_path, out_path]
+        self._driver = ElasticDriver(self.discovery, command,
+                                     min_np=self.min_np, max_np=self.max_np,
+                                     reset_limit=self.reset_limit)
+        self._driver.start()
+        err = self._driver.wait_for_result()
+        self._driver.stop()
+        try:
+            if err is not None:
+                raise err if isinstance(err, Exception) \
+                    else RuntimeError(err)
+            with open(out_path, "rb") as f:
+                return pickle.load(f)
+        finally:
+            for p in (fn_path, out_path):
+                try:
+                    os.unlink(p)
+                except OSError:
+                    pass
+
+    def shutdown(self):
+        if self._driver is not None:
+            self._driver.stop()
+            self._driver = None

@@ -78,7 +78,79 @@ def run(fn, args=(), kwargs=None, num_proc=None, use_mpi=None, use_gloo=None,
     return [r for _, r in sorted(results)]
 
 
-def run_elastic(*a, **kw):
-    raise NotImplementedError(
-        "elastic Spark jobs: use hvdrun --host-discovery-script with a "
-        "script that queries the Spark master for executor hosts")
+class SparkExecutorDiscovery:
+    """HostDiscovery backed by the SparkContext's live executor list
+    (reference: spark/runner.py run_elastic uses Spark's own view of
+    executors to drive elasticity)."""
+
+    def __init__(self, spark_context, default_slots=1):
+        self._sc = spark_context
+        self.default_slots = default_slots
+
+    def find_available_hosts_and_slots(self):
+        # executor memory status keys are "host:port"; the driver itself is
+        # excluded so workers only land on executors
+        status = self._sc._jsc.sc().getExecutorMemoryStatus()
+        keys = list(
+            self._sc._jvm.scala.collection.JavaConverters
+            .mapAsJavaMapConverter(status).asJava().keySet())
+        driver_host = self._sc._conf.get("spark.driver.host", "")
+        hosts = {}
+        for k in keys:
+            host = str(k).rsplit(":", 1)[0]
+            if host == driver_host and len(keys) > 1:
+                continue
+            hosts[host] = hosts.get(host, 0) + self.default_slots
+        return hosts
+
+
+def run_elastic(fn, args=(), kwargs=None, num_proc=None, min_np=1,
+                max_np=None, slots_per_host=1, reset_limit=None,
+                verbose=True):
+    """Elastic Spark jobs (reference: spark/runner.py:312+): the elastic
+    driver tracks Spark's live executor hosts, launches workers over ssh
+    to those hosts (the usual Spark-cluster topology), and workers run
+    `fn` under the elastic rendezvous protocol.  `fn` should use
+    hvd.elastic.run internally, exactly as with hvdrun
+    --host-discovery-script.  Requires a shared filesystem (the function
+    is shipped as a cloudpickle file, like the reference's serialized
+    train function)."""
+    import sys
+    import tempfile
+
+    import cloudpickle
+
+    _require_pyspark()
+    from pyspark.sql import SparkSession
+
+    from horovod_amd.runner.elastic_driver import ElasticDriver
+
+    spark = SparkSession.builder.getOrCreate()
+    sc = spark.sparkContext
+    discovery = SparkExecutorDiscovery(sc, default_slots=slots_per_host)
+
+    payload = cloudpickle.dumps((fn, tuple(args), dict(kwargs or {})))
+    fd, fn_path = tempfile.mkstemp(suffix=".pkl", dir=os.getcwd())
+    with os.fdopen(fd, "wb") as f:
+        f.write(payload)
+    out_path = fn_path + ".out"
+    command = [sys.executable, "-m", "horovod_amd.spark._elastic_task",
+               fn_path, out_path]
+    driver = ElasticDriver(discovery, command, min_np=min_np,
+                           max_np=max_np or num_proc,
+                           reset_limit=reset_limit)
+    driver.start()
+    err = driver.wait_for_result()
+    driver.stop()
+    try:
+        if err is not None:
+            raise err if isinstance(err, Exception) else RuntimeError(err)
+        import pickle
+        with open(out_path, "rb") as f:
+            return pickle.load(f)
+    finally:
+        for p in (fn_path, out_path):
+            try:
+                os.unlink(p)
+            except OSError:
+                pass
