@@ -236,6 +236,19 @@ def main(argv=None):
         from horovod_amd.runner.elastic_driver import run_elastic
         return run_elastic(args, command, env)
 
+    # NIC auto-discovery for multi-host launches (reference:
+    # driver_service.py probes common interfaces before launch); an explicit
+    # --network-interface always wins
+    if hosts and "NCCL_SOCKET_IFNAME" not in env:
+        from horovod_amd.runner.network import resolve_nics
+        host_names = [h.split(":")[0] for h in hosts.split(",")]
+        nics = resolve_nics(args.network_interface, host_names,
+                            verbose=args.verbose)
+        if nics:
+            env["NCCL_SOCKET_IFNAME"] = nics
+            if args.verbose:
+                print(f"[hvdrun] common NICs: {nics}")
+
     if hosts:
         codes = run_distributed(args.num_proc, hosts, command, env=env,
                                 verbose=args.verbose)
