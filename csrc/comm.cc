@@ -1,6 +1,7 @@
 #include "comm.h"
 
 #include <arpa/inet.h>
+#include <fcntl.h>
 #include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
@@ -20,6 +21,17 @@ namespace {
 void set_nodelay(int fd) {
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+}
+
+// Mesh sockets MUST be non-blocking: on Linux a blocking send() on a
+// stream socket queues the ENTIRE request in-kernel before returning, so
+// a poll loop that calls send() for the full remainder stops servicing
+// POLLIN for the duration — two ranks exchanging buffers larger than the
+// socket queues then deadlock with both tx and rx windows full (observed
+// with ~32 MB ring segments; root cause of the round-2 mesh hang).
+void set_nonblock(int fd) {
+  int flags = fcntl(fd, F_GETFL, 0);
+  if (flags >= 0) fcntl(fd, F_SETFL, flags | O_NONBLOCK);
 }
 
 [[noreturn]] void comm_error(const std::string& what) {
@@ -330,6 +342,7 @@ void MeshComm::Init(StarComm& star, const std::string& root_addr,
       left -= (size_t)n;
     }
     fds_[peer] = fd;
+    set_nonblock(fd);
   }
   int expected = size_ - 1 - rank_;
   while (expected > 0) {
@@ -354,6 +367,7 @@ void MeshComm::Init(StarComm& star, const std::string& root_addr,
     }
     if (peer <= rank_ || peer >= size_) comm_error("mesh bad peer rank");
     fds_[peer] = fd;
+    set_nonblock(fd);
     --expected;
   }
   close(listen_fd_);
@@ -408,13 +422,16 @@ void MeshComm::SendRecv2(int send_peer, const void* out, size_t out_len,
     if (si >= 0 && (p[si].revents & (POLLOUT | POLLERR | POLLHUP))) {
       ssize_t w = ::send(sfd, (const char*)out + so, out_len - so,
                          MSG_NOSIGNAL);
-      if (w <= 0) comm_error("mesh send");
-      so += (size_t)w;
+      if (w < 0 && errno != EAGAIN && errno != EWOULDBLOCK)
+        comm_error("mesh send");
+      if (w > 0) so += (size_t)w;
     }
     if (ri >= 0 && (p[ri].revents & (POLLIN | POLLERR | POLLHUP))) {
       ssize_t r = ::recv(rfd, (char*)in + ro, in_len - ro, 0);
-      if (r <= 0) comm_error("mesh recv (peer died?)");
-      ro += (size_t)r;
+      if (r == 0) comm_error("mesh recv: peer closed");
+      if (r < 0 && errno != EAGAIN && errno != EWOULDBLOCK)
+        comm_error("mesh recv (peer died?)");
+      if (r > 0) ro += (size_t)r;
     }
   }
 }

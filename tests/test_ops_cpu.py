@@ -764,3 +764,25 @@ def test_response_stream_hash_check_np3():
             ts = [torch.ones(16) for _ in range(2)]
             hvd.grouped_allreduce(ts, average=False, name=f"hg{i % 3}")
     """, extra_env={"HOROVOD_CHECK_RESPONSE_STREAM": "1"}, timeout=240)
+
+
+def test_mesh_large_transfers_np2():
+    """Round-2 regression: ring/pairwise transfers far larger than the TCP
+    socket queues (~10 MB) must not deadlock — blocking send() on Linux
+    queues the whole request before returning, so the mesh sockets have to
+    be non-blocking for the full-duplex poll loop to keep draining."""
+    run_workers(2, """
+        # 64 MB fp32 allreduce -> 32 MB ring segments each way
+        n = 16 * 1024 * 1024
+        t = torch.full((n,), float(rank + 1))
+        out = hvd.allreduce(t, average=False, name="big.ar")
+        assert out[0].item() == 3.0 and out[-1].item() == 3.0
+        # 32 MB allgather (pairwise exchange)
+        g = hvd.allgather(torch.full((n // 4, 2), float(rank)),
+                          name="big.ag")
+        assert g.shape[0] == n // 2
+        # 48 MB broadcast (binomial: pure one-directional send)
+        b = hvd.broadcast(torch.full((12 * 1024 * 1024,), 7.0),
+                          root_rank=0, name="big.bc")
+        assert b[-1].item() == 7.0
+    """, timeout=420)
