@@ -108,6 +108,10 @@ void ResponseCache::EvictSet(int32_t process_set_id) {
       Evict(i);
 }
 
+void ResponseCache::Clear() {
+  for (int i = 0; i < (int)entries_.size(); ++i) Evict(i);
+}
+
 void ResponseCache::Evict(int slot) {
   Entry& e = entries_[slot];
   if (!e.live) return;
@@ -205,6 +209,10 @@ Controller::Controller(StarComm* comm, int rank, int size, ControllerConfig cfg)
     : comm_(comm), rank_(rank), size_(size), cfg_(cfg) {
   cache_.set_capacity(cfg_.cache_capacity);
   check_stream_ = std::getenv("HOROVOD_CHECK_RESPONSE_STREAM") != nullptr;
+  if (const char* tp = std::getenv("HOROVOD_TRACE_CYCLES")) {
+    std::string path = std::string(tp) + "." + std::to_string(rank_);
+    trace_ = std::fopen(path.c_str(), "w");
+  }
   start_time_ = std::chrono::steady_clock::now();
   if (rank_ == 0 && size_ > 1 && std::getenv("HOROVOD_AUTOTUNE")) {
     const char* lp = std::getenv("HOROVOD_AUTOTUNE_LOG");
@@ -266,9 +274,13 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
   }
 
   // re-classify pending hits whose cache slot was since evicted/replaced
-  // (capacity-pressure Put evictions would otherwise strand them forever)
+  // (capacity-pressure Put evictions would otherwise strand them forever).
+  // A cache-off TUNE also drains ALL pending to the slow path: a request
+  // classified HIT just before the flip would otherwise keep fast-path
+  // voting while its peers renegotiate — cross-rank deadlock.
   for (auto it = cached_pending_.begin(); it != cached_pending_.end();) {
-    if (cache_.Lookup(*it) != ResponseCache::State::HIT) {
+    if (!cache_enabled_ ||
+        cache_.Lookup(*it) != ResponseCache::State::HIT) {
       slow.push_back(*it);
       it = cached_pending_.erase(it);
     } else {
@@ -444,6 +456,19 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
     for (auto& r : fused) result.responses.push_back(std::move(r));
   }
 
+  if (trace_) {
+    std::fprintf(trace_, "c%llu new=%zu slow_held=%zu pend=%zu resp=",
+                 (unsigned long long)trace_cycle_++, new_requests.size(),
+                 inflight_.size(), cached_pending_.size());
+    for (auto& r : result.responses) {
+      std::fprintf(trace_, "%d:", (int)r.type);
+      for (auto& n : r.names) std::fprintf(trace_, "%s,", n.c_str());
+      std::fprintf(trace_, ";");
+    }
+    std::fprintf(trace_, " cache=%d fus=%lld cyc=%.3f\n", (int)cache_enabled_,
+                 (long long)cfg_.fusion_threshold_bytes, cfg_.cycle_time_ms);
+    std::fflush(trace_);
+  }
   if (check_stream_) {
     // FNV-1a over each response's wire serialization, in execution order
     for (auto& resp : result.responses) {

@@ -76,6 +76,11 @@ class ResponseCache {
            int32_t my_local_index = -1);
   void Evict(int slot);
   void EvictSet(int32_t process_set_id);
+  // Evict everything (cache-off TUNE: the reference clears on disable so a
+  // later re-enable rebuilds every rank's state from scratch — stale live
+  // slots would otherwise classify HIT on one rank while a peer's copy is
+  // already in flight on the slow path: cross-rank deadlock).
+  void Clear();
   size_t num_slots() const { return entries_.size(); }
   bool slot_live(int slot) const { return entries_[slot].live; }
 
@@ -144,6 +149,7 @@ class Controller {
                  bool cache_enabled = true) {
     cfg_.fusion_threshold_bytes = fusion_bytes;
     cfg_.cycle_time_ms = cycle_time_ms;
+    if (cache_enabled_ && !cache_enabled) cache_.Clear();
     cache_enabled_ = cache_enabled;
   }
 
@@ -190,6 +196,9 @@ class Controller {
   // diverging fast path before it deadlocks RCCL
   bool check_stream_ = false;
   uint64_t response_hash_ = 1469598103934665603ull;
+  // HOROVOD_TRACE_CYCLES=<prefix>: per-cycle debug log (prefix.<rank>)
+  std::FILE* trace_ = nullptr;
+  uint64_t trace_cycle_ = 0;
 
  public:
   // join bookkeeping shared with core
