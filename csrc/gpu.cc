@@ -102,6 +102,13 @@ struct DeviceCtx {
   at::Tensor dots_buffer;      // adasum per-tensor {dot,|a|2,|b|2} doubles
   std::unordered_map<int32_t, ncclComm_t> comms;  // process_set -> comm
   std::unordered_map<int32_t, OneshotCtx> oneshot;  // process_set -> window
+  // hierarchical/torus allreduce sub-communicators (global set only)
+  struct HierCtx {
+    ncclComm_t local = nullptr;  // ranks on my node
+    ncclComm_t cross = nullptr;  // same local_rank across nodes
+    bool ready = false;
+  };
+  std::unordered_map<int32_t, HierCtx> hier;
 
   explicit DeviceCtx(int dev)
       : device(dev),
@@ -342,6 +349,11 @@ void AbortAllCommsLocked(const char* why) {
   for (auto& kv : g_ctx) {
     for (auto& ck : kv.second->comms) (void)ncclCommAbort(ck.second);
     kv.second->comms.clear();
+    for (auto& hk : kv.second->hier) {
+      if (hk.second.local) (void)ncclCommAbort(hk.second.local);
+      if (hk.second.cross) (void)ncclCommAbort(hk.second.cross);
+    }
+    kv.second->hier.clear();
     // raise the one-shot abort word (flags[3]) so spin-wait kernels exit
     // instead of wedging the GPU on a dead peer (SDMA write proceeds even
     // while compute is spinning)
@@ -397,6 +409,8 @@ ncclComm_t EnsureComm(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
 std::unordered_map<int32_t, bool> g_bootstrapped;
 // Same for the one-shot shared-memory windows.
 std::unordered_map<int32_t, bool> g_oneshot_bootstrapped;
+// ...and for the hierarchical sub-comms.
+std::unordered_map<int32_t, bool> g_hier_bootstrapped;
 
 // ---- one-shot allreduce host side -----------------------------------------
 
@@ -469,6 +483,29 @@ int OneshotOpCode(ReduceOp op) {
     case ReduceOp::PRODUCT: return 3;
     default: return 0;  // sum / average
   }
+}
+
+// ---- hierarchical / torus allreduce ---------------------------------------
+// Two-level schedule for multi-node jobs (reference: NCCLHierarchicalAllreduce
+// nccl_operations.cc:307-577 and the non-upstream NCCLTorusAllreduce 606-829):
+// intra-node ncclReduceScatter -> cross-node ncclAllReduce on the shard (one
+// GPU per node column) -> intra-node ncclAllGather.  All-RCCL (the reference
+// hops through host MPI for the cross step; RCCL's own transports cover it).
+// Enabled with HOROVOD_HIERARCHICAL_ALLREDUCE=1 when cross_size > 1; a
+// single-node test can fake a 2x4 topology via the HOROVOD_LOCAL_* env.
+
+bool HierEnabled() {
+  static bool v = std::getenv("HOROVOD_HIERARCHICAL_ALLREDUCE") != nullptr;
+  return v;
+}
+
+bool HierEligible(GlobalState& st, const Response& resp, int n_set) {
+  if (!HierEnabled() || resp.type != ResponseType::ALLREDUCE) return false;
+  if (n_set != st.size || st.local_size <= 1 || st.cross_size <= 1)
+    return false;
+  // contiguous rank layout (rank = cross_rank*local_size + local_rank) is
+  // what the launchers produce; anything else falls back to flat
+  return st.rank == st.cross_rank * st.local_size + st.local_rank;
 }
 
 // All-rank string exchange over the star (every global rank participates).
@@ -724,6 +761,42 @@ int64_t PackEntries(DeviceCtx& ctx, std::vector<TensorTableEntry>& entries,
   return total;
 }
 
+
+// Create the local + cross RCCL comms for the hierarchical schedule.
+// Lock-step over the star: every rank participates in one id-exchange
+// round (local leaders contribute the local id, node-0 ranks the cross id
+// for their column), then all ranks init local comms, then cross comms.
+bool EnsureHier(GlobalState& st, DeviceCtx& ctx, int32_t set_id) {
+  auto& h = ctx.hier[set_id];
+  if (h.ready) return true;
+  int ls = st.local_size, cs = st.cross_size, lr = st.local_rank,
+      cr = st.cross_rank;
+  ncclUniqueId lid{}, cid{};
+  std::string payload(2 * sizeof(ncclUniqueId), '\0');
+  if (lr == 0) {
+    RCCL_CHECK(ncclGetUniqueId(&lid));
+    std::memcpy(&payload[0], &lid, sizeof(lid));
+  }
+  if (cr == 0) {
+    RCCL_CHECK(ncclGetUniqueId(&cid));
+    std::memcpy(&payload[sizeof(lid)], &cid, sizeof(cid));
+  }
+  auto all = ExchangeAllGather(st, payload);
+  int local_leader = cr * ls;          // local_rank 0 on my node
+  int cross_leader = lr;               // my column's rank on node 0
+  if (local_leader >= (int)all.size() || cross_leader >= (int)all.size() ||
+      all[local_leader].size() < 2 * sizeof(ncclUniqueId) ||
+      all[cross_leader].size() < 2 * sizeof(ncclUniqueId))
+    return false;
+  std::memcpy(&lid, all[local_leader].data(), sizeof(lid));
+  std::memcpy(&cid, all[cross_leader].data() + sizeof(lid), sizeof(cid));
+  c10::hip::HIPGuard guard(ctx.device);
+  RCCL_CHECK(ncclCommInitRank(&h.local, ls, lid, lr));
+  RCCL_CHECK(ncclCommInitRank(&h.cross, cs, cid, cr));
+  h.ready = true;
+  HVD_LOG(INFO, "hierarchical comms ready: %dx%d (local x cross)", ls, cs);
+  return true;
+}
 
 // ---- Distributed VHDD Adasum (reference: adasum_gpu_operations.cc:44-120
 // structure + adasum.h:195-345 FusedAllreduce) -----------------------------
@@ -1092,6 +1165,18 @@ void Execute(GlobalState& st, Response& resp,
     if (!ok && !g_bootstrapped.count(sid)) rccl_bootstrap();
   }
   if (!oneshot_resp && !g_bootstrapped.count(sid)) rccl_bootstrap();
+  // hierarchical comms piggyback their one id-exchange round on the first
+  // eligible response (global set => every rank is a member and reaches
+  // this in lock-step)
+  bool hier_resp = !oneshot_resp && HierEligible(st, resp, (int)set.ranks.size());
+  if (hier_resp && !g_hier_bootstrapped.count(sid)) {
+    if (!entries.empty()) {
+      EnsureHier(st, GetCtx(entries[0].device), sid);
+    } else {
+      ExchangeAllGather(st, "");  // relay (unreachable for global sets)
+    }
+    g_hier_bootstrapped[sid] = true;
+  }
   if (entries.empty()) return;  // relay-only rank (not a member)
   if (entries.size() != resp.names.size())
     throw std::runtime_error(
@@ -1165,6 +1250,41 @@ void Execute(GlobalState& st, Response& resp,
         HIP_CHECK(OneshotReduceLaunch(
             a, os->seq, (int)wire, (unsigned long long)(total * wire_size),
             fb.data_ptr(), OneshotOpCode(resp.reduce_op), stream));
+        for (auto& e : entries)
+          if (!e.output.defined()) {
+            e.output = at::empty_like(e.tensor);
+            RecordStreamFor(e.output, ctx.stream);
+          }
+        PackEntries(ctx, entries, wire, true);
+        break;
+      }
+      DeviceCtx::HierCtx* hc = nullptr;
+      if (hier_resp) {
+        auto hit = ctx.hier.find(sid);
+        if (hit != ctx.hier.end() && hit->second.ready) hc = &hit->second;
+      }
+      if (hc) {
+        // two-level: local RS -> cross AR on the shard -> local AG
+        activity = "HIER_ALLREDUCE";
+        int64_t total = PackEntries(ctx, entries, wire, false);
+        char* base = (char*)ctx.fusion_buffer.data_ptr();
+        int ls = st.local_size, lr = st.local_rank;
+        if (total > 0 && total % ls == 0) {
+          int64_t shard = total / ls;
+          char* mine = base + (int64_t)lr * shard * wire_size;
+          RCCL_CHECK(ncclReduceScatter(base, mine, shard, wire_nccl,
+                                       ToNcclOp(resp.reduce_op), hc->local,
+                                       stream));
+          RCCL_CHECK(ncclAllReduce(mine, mine, shard, wire_nccl,
+                                   ToNcclOp(resp.reduce_op), hc->cross,
+                                   stream));
+          RCCL_CHECK(ncclAllGather(mine, base, shard, wire_nccl, hc->local,
+                                   stream));
+        } else {
+          // fused length not divisible by local_size: flat fallback
+          RCCL_CHECK(ncclAllReduce(base, base, total, wire_nccl,
+                                   ToNcclOp(resp.reduce_op), comm, stream));
+        }
         for (auto& e : entries)
           if (!e.output.defined()) {
             e.output = at::empty_like(e.tensor);
@@ -1561,6 +1681,12 @@ void Shutdown() {
         if (ok.second.my_staging) (void)hipFree(ok.second.my_staging);
         if (ok.second.my_flags) (void)hipFree(ok.second.my_flags);
       }
+      for (auto& hk : kv.second->hier) {
+        if (!g_comm_failed) {
+          if (hk.second.local) ncclCommDestroy(hk.second.local);
+          if (hk.second.cross) ncclCommDestroy(hk.second.cross);
+        }
+      }
       delete kv.second;
     }
     g_ctx.clear();
@@ -1568,6 +1694,7 @@ void Shutdown() {
   g_comm_failed = false;  // elastic re-init starts clean
   g_bootstrapped.clear();
   g_oneshot_bootstrapped.clear();
+  g_hier_bootstrapped.clear();
   std::lock_guard<std::mutex> g(g_event_mu);
   for (auto ev : g_event_pool) (void)hipEventDestroy(ev);
   g_event_pool.clear();

@@ -412,3 +412,57 @@ def test_multi_adasum_rsvhdd(np_):
                              name="rsv2")
         assert torch.equal(out, out2)
     """, timeout=420)
+
+
+@requires_multi_gpu
+def test_multi_hierarchical_allreduce():
+    """Hierarchical/torus allreduce (local RS -> cross AR -> local AG) on a
+    FAKED 2-node topology: np GPUs presented as 2 nodes x np/2 — the same
+    RCCL calls a real multi-node job issues, minus the network hop."""
+    import os
+    import subprocess
+    import sys
+    from horovod_amd.runner.launch import find_free_port, slot_env
+    from tests.parallel_util import REPO
+
+    np_ = min(NGPU, 8)
+    if np_ % 2:
+        np_ -= 1
+    ls = np_ // 2  # slots per fake node
+    body = (
+        "import torch, horovod_amd.torch as hvd\n"
+        "hvd.init(); torch.cuda.set_device(int(__import__('os').environ['TEST_DEV']))\n"
+        "d = torch.device('cuda', int(__import__('os').environ['TEST_DEV']))\n"
+        "n = hvd.size(); tri = n * (n + 1) // 2\n"
+        "for i in range(6):\n"
+        "    t = torch.full((4096,), float(hvd.rank() + 1), device=d)\n"
+        "    o = hvd.allreduce(t, average=False, name=f'h{i}')\n"
+        "    assert torch.equal(o, torch.full_like(t, float(tri))), i\n"
+        "    ts = [torch.full((257,), float(hvd.rank() + 1), device=d)\n"
+        "          for _ in range(3)]\n"
+        "    outs = hvd.grouped_allreduce(ts, average=False, name=f'hg{i}')\n"
+        "    for o2 in outs:\n"
+        "        assert torch.allclose(o2, torch.full_like(o2, float(tri)))\n"
+        "print('HIER_OK', hvd.rank())\n"
+    )
+    port = find_free_port()
+    procs = []
+    for r in range(np_):
+        env = slot_env(r, np_, r % ls, ls, r // ls, 2, "127.0.0.1", port)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        env["HOROVOD_HIERARCHICAL_ALLREDUCE"] = "1"
+        env["TEST_DEV"] = str(r)  # real device != fake local_rank
+        procs.append(subprocess.Popen([sys.executable, "-c", body], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, text=True))
+    outs = []
+    for r, p in enumerate(procs):
+        try:
+            out, _ = p.communicate(timeout=300)
+        except subprocess.TimeoutExpired:
+            for q in procs:
+                q.kill()
+            raise AssertionError(f"rank {r} timed out (hierarchical)")
+        outs.append(out)
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0 and "HIER_OK" in out, (r, out[-500:])
