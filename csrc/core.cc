@@ -322,6 +322,145 @@ void MeshRingAllreduce(GlobalState& st, const ProcessSetInfo& set,
   }
 }
 
+// Distributed VHDD Adasum over the mesh (CPU twin of gpu.cc's
+// ExecuteAdasumRSVHDD; reference: adasum_mpi.cc point-to-point VHDD).
+// Works entirely within set MEMBERS (scalar group sums ride the mesh, not
+// the star), so subset Adasum is supported.  All math in float64, same
+// combine tree as the AdasumCombineInPlace golden.
+void MeshAdasum(GlobalState& st, const ProcessSetInfo& set,
+                const Response& resp,
+                std::vector<TensorTableEntry>& entries) {
+  const int n = (int)set.ranks.size();
+  const int li = set.local_index(st.rank);
+  DataType wire = resp.dtype;
+  std::vector<int64_t> offsets, counts;
+  PerTensorLayout(resp, offsets, counts);
+  const int64_t T = (int64_t)counts.size();
+
+  std::vector<at::Tensor> flats;
+  for (auto& e : entries) flats.push_back(FlatPrescaled(e, wire));
+  at::Tensor work =
+      (flats.size() == 1 ? flats[0] : at::cat(flats)).to(at::kDouble);
+  if (!work.is_contiguous()) work = work.contiguous();
+  const int64_t L = work.numel();
+  double* wp = work.data_ptr<double>();
+
+  int p = 1;
+  while (p * 2 <= n) p *= 2;
+  int levels = 0;
+  while ((1 << levels) < p) ++levels;
+
+  // fold the non-power-of-2 remainder: (i-p, i) pairs, full-vector combine
+  if (li >= p) {
+    st.mesh.Send(set.ranks[li - p], wp, (size_t)(L * 8));
+  } else if (li + p < n) {
+    at::Tensor peer = at::empty({L}, work.options());
+    st.mesh.Recv(set.ranks[li + p], peer.data_ptr(), (size_t)(L * 8));
+    AdasumCombineInPlace(work, peer, offsets, counts);
+  }
+
+  // per-owner final piece ranges are a pure function of the split path
+  auto final_range = [&](int r) {
+    int64_t S = 0, len = L;
+    for (int k = 0; k < levels; ++k) {
+      int64_t half = len / 2;
+      if ((r >> k) & 1) {
+        S += half;
+        len -= half;
+      } else {
+        len = half;
+      }
+    }
+    return std::pair<int64_t, int64_t>(S, len);
+  };
+
+  int64_t S = 0, len = L;
+  if (li < p) {
+    at::Tensor scratch = at::empty({(L + 1) / 2}, work.options());
+    for (int k = 0; k < levels; ++k) {
+      int stride = 1 << k;
+      int partner = li ^ stride;
+      bool lower = li < partner;
+      int64_t half = len / 2;
+      int64_t my_len = lower ? half : len - half;
+      int64_t myS = lower ? S : S + half;
+      int64_t th_len = len - my_len;
+      int64_t thS = lower ? S + half : S;
+      st.mesh.SendRecv2(set.ranks[partner], wp + thS, (size_t)(th_len * 8),
+                        set.ranks[partner], scratch.data_ptr(),
+                        (size_t)(my_len * 8));
+      double* sp = scratch.data_ptr<double>();
+      // canonical per-tensor partial dots over my piece (A = lower's data)
+      std::vector<double> part((size_t)(3 * T), 0.0);
+      for (int64_t t = 0; t < T; ++t) {
+        int64_t lo = std::max(myS, offsets[t]);
+        int64_t hi = std::min(myS + my_len, offsets[t] + counts[t]);
+        for (int64_t i = lo; i < hi; ++i) {
+          double mine = wp[i], theirs = sp[i - myS];
+          double a = lower ? mine : theirs, b = lower ? theirs : mine;
+          part[t * 3 + 0] += a * b;
+          part[t * 3 + 1] += a * a;
+          part[t * 3 + 2] += b * b;
+        }
+      }
+      // group-sum the scalars over the 2*stride ranks sharing this vector
+      // (pairwise mesh exchange; sum in ascending member order so every
+      // rank derives bit-identical coefficients)
+      int group = 2 * stride;
+      int g0 = (li / group) * group;
+      std::vector<std::vector<double>> by_member((size_t)group);
+      by_member[li - g0] = part;
+      std::vector<double> inbox((size_t)(3 * T));
+      for (int j = 1; j < group; ++j) {
+        // XOR pairing: symmetric (x^j partners with x), so the same-peer
+        // full-duplex exchange is matched on both sides (group is a
+        // power of 2)
+        int peer_li = g0 + (((li - g0) ^ j) & (group - 1));
+        st.mesh.SendRecv2(set.ranks[peer_li], part.data(),
+                          part.size() * 8, set.ranks[peer_li], inbox.data(),
+                          inbox.size() * 8);
+        by_member[peer_li - g0] = inbox;
+      }
+      std::vector<double> dots((size_t)(3 * T), 0.0);
+      for (int m = 0; m < group; ++m)
+        for (size_t i = 0; i < dots.size(); ++i)
+          dots[i] += by_member[m].empty() ? 0.0 : by_member[m][i];
+      // combine my piece: result = acoef*A + bcoef*B
+      for (int64_t t = 0; t < T; ++t) {
+        double dot = dots[t * 3], na = dots[t * 3 + 1], nb = dots[t * 3 + 2];
+        double ac = na > 0 ? 1.0 - dot / (2.0 * na) : 1.0;
+        double bc = nb > 0 ? 1.0 - dot / (2.0 * nb) : 1.0;
+        double cm = lower ? ac : bc;  // coefficient of MY data
+        double cr = lower ? bc : ac;  // coefficient of the received data
+        int64_t lo = std::max(myS, offsets[t]);
+        int64_t hi = std::min(myS + my_len, offsets[t] + counts[t]);
+        for (int64_t i = lo; i < hi; ++i)
+          wp[i] = cm * wp[i] + cr * sp[i - myS];
+      }
+      S = myS;
+      len = my_len;
+    }
+  }
+
+  // regather: each of the p owners ships its final piece to every other
+  // member (folded ranks included), pair transfers in fixed (r, dst) order
+  for (int r = 0; r < p; ++r) {
+    auto [Sr, lr] = final_range(r);
+    if (lr == 0) continue;
+    if (li == r) {
+      for (int dst = 0; dst < n; ++dst)
+        if (dst != li)
+          st.mesh.Send(set.ranks[dst], wp + Sr, (size_t)(lr * 8));
+    } else {
+      st.mesh.Recv(set.ranks[r], wp + Sr, (size_t)(lr * 8));
+    }
+  }
+
+  at::Tensor out = work.to(DataTypeToTorch(wire));
+  for (size_t i = 0; i < entries.size(); ++i)
+    UnpackInto(entries[i], out.narrow(0, offsets[i], counts[i]));
+}
+
 void MeshAllreduce(GlobalState& st, const ProcessSetInfo& set,
                    const Response& resp,
                    std::vector<TensorTableEntry>& entries) {
@@ -536,9 +675,14 @@ void MeshReducescatter(GlobalState& st, const ProcessSetInfo& set,
 void CPUAllreduce(GlobalState& st, const Response& resp,
                   std::vector<TensorTableEntry>& entries, bool member) {
   auto& set = st.controller->process_set(resp.process_set_id);
-  if (resp.type != ResponseType::ADASUM && UseMesh(st, set)) {
-    // ring over direct links; non-members do nothing (no star frames)
-    if (member && !entries.empty()) MeshAllreduce(st, set, resp, entries);
+  if (UseMesh(st, set)) {
+    // direct links; non-members do nothing (no star frames)
+    if (member && !entries.empty()) {
+      if (resp.type == ResponseType::ADASUM)
+        MeshAdasum(st, set, resp, entries);
+      else
+        MeshAllreduce(st, set, resp, entries);
+    }
     return;
   }
   DataType wire = resp.dtype;

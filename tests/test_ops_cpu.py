@@ -786,3 +786,27 @@ def test_mesh_large_transfers_np2():
                           root_rank=0, name="big.bc")
         assert b[-1].item() == 7.0
     """, timeout=420)
+
+
+def test_subset_adasum_np4():
+    """Round-2: CPU Adasum runs distributed VHDD over mesh links entirely
+    within the subset's members (scalar sums no longer need the star, so
+    subset Adasum works with non-members running free)."""
+    run_workers(4, """
+        ps = hvd.add_process_set(hvd.ProcessSet([1, 2]))
+        if rank in (1, 2):
+            def vec(r):
+                return torch.cos(torch.arange(37, dtype=torch.float64) *
+                                 (r + 2))
+            def combine(a, b):
+                dot = (a*b).sum(); na = (a*a).sum(); nb = (b*b).sum()
+                return a*(1 - dot/(2*na)) + b*(1 - dot/(2*nb))
+            expected = combine(vec(1), vec(2)).float()
+            out = hvd.allreduce(vec(rank).float(), op=hvd.Adasum,
+                                name="sada", process_set=ps)
+            assert torch.allclose(out, expected, atol=1e-6), \
+                (out - expected).abs().max()
+        # everyone still agrees on a global op afterwards
+        g = hvd.allreduce(torch.ones(4), average=False, name="after")
+        assert g[0].item() == 4.0
+    """, timeout=240)
