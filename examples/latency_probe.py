@@ -53,3 +53,32 @@ def tiny_sync():
 
 timed("torch.distributed tiny + synchronize", tiny_sync)
 hvd.shutdown()
+
+
+# ---- multi-rank crossover sweep ------------------------------------------
+# Run on a multi-GPU box:
+#   bin/hvdrun -np 8 python examples/latency_probe.py                 # ring
+#   HOROVOD_ONESHOT_ALLREDUCE=1 bin/hvdrun -np 8 python examples/...  # one-shot
+# and compare the per-size rows: the crossover where ring beats one-shot is
+# the value to hand HOROVOD_ONESHOT_THRESHOLD (the autotuner explores it
+# too — docs/autotune.md).
+if hvd.size() > 1:
+    dev = torch.device("cuda", hvd.local_rank())
+    algo = "one-shot<=%s" % os.environ.get("HOROVOD_ONESHOT_THRESHOLD",
+                                           "4MiB") \
+        if os.environ.get("HOROVOD_ONESHOT_ALLREDUCE") else "rccl-ring"
+    if hvd.rank() == 0:
+        print(f"\n-- np={hvd.size()} bucket sweep ({algo}) --", flush=True)
+    for numel in (1 << 10, 1 << 14, 1 << 17, 1 << 20, 1 << 22):
+        t = torch.ones(numel, device=dev)
+        label = f"allreduce {numel * 4 // 1024:>8d} KB x np{hvd.size()}"
+        for _ in range(5):
+            hvd.allreduce(t, average=False, name="sw")
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(30):
+            hvd.allreduce(t, average=False, name="sw")
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 30 * 1e3
+        if hvd.rank() == 0:
+            print(f"{label:46s} {dt:8.3f} ms/op", flush=True)
