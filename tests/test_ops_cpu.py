@@ -810,3 +810,41 @@ def test_subset_adasum_np4():
         g = hvd.allreduce(torch.ones(4), average=False, name="after")
         assert g[0].item() == 4.0
     """, timeout=240)
+
+
+def test_error_mismatched_dtype_np2():
+    """Mismatched dtypes across ranks produce the reference's named error
+    on every rank (reference controller.cc:599-602 error text)."""
+    run_workers(2, """
+        t = torch.ones(4) if rank == 0 else torch.ones(4, dtype=torch.float64)
+        try:
+            hvd.allreduce(t, average=False, name="baddt")
+            raise SystemExit("expected a dtype-mismatch error")
+        except RuntimeError as e:
+            assert "Mismatched data types" in str(e), e
+        # the job survives the error: a good op still completes
+        out = hvd.allreduce(torch.ones(4), average=False, name="gooddt")
+        assert out[0].item() == 2.0
+    """)
+
+
+def test_error_mismatched_root_np2():
+    run_workers(2, """
+        try:
+            hvd.broadcast(torch.ones(4), root_rank=rank, name="badroot")
+            raise SystemExit("expected a root-mismatch error")
+        except RuntimeError as e:
+            assert "root rank" in str(e).lower(), e
+    """)
+
+
+def test_error_mismatched_reduce_op_np2():
+    run_workers(2, """
+        op = hvd.Sum if rank == 0 else hvd.Max
+        try:
+            hvd.allreduce(torch.ones(4), op=op, name="badop")
+            raise SystemExit("expected a reduce-op-mismatch error")
+        except RuntimeError as e:
+            assert "reduction op" in str(e).lower() or \
+                "Mismatched" in str(e), e
+    """)
