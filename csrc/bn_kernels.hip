@@ -277,6 +277,36 @@ int grid_for(long long total, int C) {
     default: return hipErrorInvalidValue;                         \
   }
 
+__global__ __launch_bounds__(256) void bn_finalize_k(
+    const float* __restrict__ sums, const float* __restrict__ sqs,
+    float* __restrict__ mean, float* __restrict__ invstd,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    long long count, float momentum, float eps, int C) {
+  const float inv_n = 1.0f / (float)count;
+  const float ub = count > 1 ? (float)count / (float)(count - 1) : 1.0f;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float m = sums[c] * inv_n;
+    float v = sqs[c] * inv_n - m * m;
+    mean[c] = m;
+    invstd[c] = rsqrtf(v + eps);
+    if (running_mean) {
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+      running_var[c] =
+          (1.f - momentum) * running_var[c] + momentum * v * ub;
+    }
+  }
+}
+
+hipError_t BnFinalizeLaunch(const float* sums, const float* sqs, float* mean,
+                            float* invstd, float* running_mean,
+                            float* running_var, long long count,
+                            float momentum, float eps, int C,
+                            hipStream_t stream) {
+  bn_finalize_k<<<1, 256, 0, stream>>>(sums, sqs, mean, invstd, running_mean,
+                                       running_var, count, momentum, eps, C);
+  return hipGetLastError();
+}
+
 hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
                          float* sums, float* sqs, hipStream_t stream) {
   int blocks = grid_for(total, C);
