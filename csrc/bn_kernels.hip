@@ -254,8 +254,15 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_k(
   }
 }
 
-int grid_for(long long total, int C) {
-  long long blocks = (total / 8 + 255) / 256;
+int grid_for(long long total, int C, int vecs_per_thread = 1) {
+  // vecs_per_thread > 1 for the REDUCTION kernels: their per-block flush
+  // costs 2*C global atomicAdds, so wide-channel shapes with few rows must
+  // amortize it over more elements per block (measured: at C=2048, N*H*W
+  // = 3136 the old 1-vec grid issued ~1.3 atomics PER ELEMENT and lost
+  // ~40% to MIOpen; 8 vecs/thread cuts the flush traffic 8x).  Elementwise
+  // kernels keep 1 for maximum bandwidth-filling parallelism.
+  long long blocks = (total / 8 + 256LL * vecs_per_thread - 1) /
+                     (256LL * vecs_per_thread);
   blocks = std::min<long long>(blocks > 0 ? blocks : 1, 2048);
   // fixed-channel decomposition invariant: (blocks*256*8) % C == 0 so each
   // thread's channel octet is stride-invariant.  vecs_per_row = C/8 <= 512;
@@ -309,7 +316,7 @@ hipError_t BnFinalizeLaunch(const float* sums, const float* sqs, float* mean,
 
 hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
                          float* sums, float* sqs, hipStream_t stream) {
-  int blocks = grid_for(total, C);
+  int blocks = grid_for(total, C, 8);
   size_t lds = 2 * (size_t)C * sizeof(float);
   DISPATCH_T(dt, (bn_stats_k<scalar_t><<<blocks, 256, lds, stream>>>(
                      (const scalar_t*)x, total, C, sums, sqs)));
@@ -340,7 +347,7 @@ hipError_t BnBwdStatsLaunch(const void* x, const void* y, const void* dy,
                             const float* invstd, long long total, int C,
                             int dt, float* sum_g, float* sum_gx,
                             hipStream_t stream) {
-  int blocks = grid_for(total, C);
+  int blocks = grid_for(total, C, 8);
   size_t lds = 2 * (size_t)C * sizeof(float);
   if (g_out) {
     DISPATCH_T(dt,
