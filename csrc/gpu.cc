@@ -1589,22 +1589,28 @@ std::vector<at::Tensor> FusedBnReluForward(at::Tensor x, at::Tensor residual,
   // (mean/invstd + running-stat update) runs in bn_finalize_k — the
   // previous ~10 tiny ATen launches per BN layer cost 3.5 ms/step of pure
   // launch overhead at batch 64 (53 layers, fwd+bwd)
-  at::Tensor ws = at::empty({4, (int64_t)C}, fopts);
+  const int64_t bank_elems = (int64_t)kBnBanks * 2 * C;
+  at::Tensor ws = at::empty({bank_elems + 2 * C}, fopts);
   float* wsp = ws.data_ptr<float>();
-  HIP_CHECK(hipMemsetAsync(wsp, 0, 2 * (size_t)C * sizeof(float), stream));
-  HIP_CHECK(BnStatsLaunch(x.data_ptr(), total, (int)C, dt, wsp, wsp + C,
+  float* mean_p = wsp + bank_elems;
+  float* invstd_p = mean_p + C;
+  HIP_CHECK(hipMemsetAsync(wsp, 0, (size_t)bank_elems * sizeof(float),
+                           stream));
+  HIP_CHECK(BnStatsLaunch(x.data_ptr(), total, (int)C, dt, wsp, nullptr,
                           stream));
+  HIP_CHECK(BnBankReduceLaunch(wsp, (int)C, stream));
   bool stats_f32 = running_mean.defined() &&
                    running_mean.scalar_type() == at::kFloat &&
                    running_mean.is_contiguous() &&
                    running_var.scalar_type() == at::kFloat &&
                    running_var.is_contiguous();
   HIP_CHECK(BnFinalizeLaunch(
-      wsp, wsp + C, wsp + 2 * C, wsp + 3 * C,
+      wsp, mean_p, invstd_p,
       stats_f32 ? running_mean.data_ptr<float>() : nullptr,
       stats_f32 ? running_var.data_ptr<float>() : nullptr, count,
       (float)momentum, (float)eps, (int)C, stream));
-  at::Tensor mean = ws[2], invstd = ws[3];  // views of the workspace
+  at::Tensor mean = ws.narrow(0, bank_elems, C);
+  at::Tensor invstd = ws.narrow(0, bank_elems + C, C);
   if (!stats_f32 && running_mean.defined()) {
     // rare non-fp32 running stats: fall back to ATen for the update only
     at::Tensor var = at::pow(invstd, -2) - eps;
@@ -1625,7 +1631,7 @@ std::vector<at::Tensor> FusedBnReluForward(at::Tensor x, at::Tensor residual,
   at::Tensor y = at::empty_like(x);
   HIP_CHECK(BnApplyReluLaunch(
       x.data_ptr(), residual.defined() ? residual.data_ptr() : nullptr,
-      y.data_ptr(), wsp + 2 * C, wsp + 3 * C, gamma_f.data_ptr<float>(),
+      y.data_ptr(), mean_p, invstd_p, gamma_f.data_ptr<float>(),
       beta_f.data_ptr<float>(), total, (int)C, dt, stream));
   return {y, mean, invstd};
 }
@@ -1646,16 +1652,17 @@ std::vector<at::Tensor> FusedBnReluBackward(at::Tensor x, at::Tensor y,
     dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
 
   auto fopts = at::TensorOptions().dtype(at::kFloat).device(x.device());
-  at::Tensor sums = at::empty({2, C}, fopts);
-  HIP_CHECK(hipMemsetAsync(sums.data_ptr<float>(), 0,
-                           2 * (size_t)C * sizeof(float), stream));
+  at::Tensor sums = at::empty({(int64_t)kBnBanks * 2 * C}, fopts);
+  float* bankp = sums.data_ptr<float>();
+  HIP_CHECK(hipMemsetAsync(bankp, 0,
+                           (size_t)kBnBanks * 2 * C * sizeof(float), stream));
   at::Tensor dres;
   if (need_residual_grad) dres = at::empty_like(x);
   HIP_CHECK(BnBwdStatsLaunch(x.data_ptr(), y.data_ptr(), dy.data_ptr(),
                              need_residual_grad ? dres.data_ptr() : nullptr,
                              mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                             total, (int)C, dt, sums[0].data_ptr<float>(),
-                             sums[1].data_ptr<float>(), stream));
+                             total, (int)C, dt, bankp, nullptr, stream));
+  HIP_CHECK(BnBankReduceLaunch(bankp, (int)C, stream));
   at::Tensor gamma_f =
       gamma.scalar_type() == at::kFloat && gamma.is_contiguous()
           ? gamma
@@ -1664,14 +1671,15 @@ std::vector<at::Tensor> FusedBnReluBackward(at::Tensor x, at::Tensor y,
   HIP_CHECK(BnBwdApplyLaunch(x.data_ptr(), y.data_ptr(), dy.data_ptr(),
                              dx.data_ptr(), mean.data_ptr<float>(),
                              invstd.data_ptr<float>(), gamma_f.data_ptr<float>(),
-                             sums[0].data_ptr<float>(), sums[1].data_ptr<float>(),
-                             total, (int)C, dt, (float)(1.0 / count), stream));
+                             bankp, bankp + C, total, (int)C, dt,
+                             (float)(1.0 / count), stream));
+  at::Tensor sum_g = sums.narrow(0, 0, C), sum_gx = sums.narrow(0, C, C);
   at::Tensor dbeta = gamma.scalar_type() == at::kFloat
-                         ? sums[0]
-                         : sums[0].to(gamma.scalar_type());
+                         ? sum_g
+                         : sum_g.to(gamma.scalar_type());
   at::Tensor dgamma = gamma.scalar_type() == at::kFloat
-                          ? sums[1]
-                          : sums[1].to(gamma.scalar_type());
+                          ? sum_gx
+                          : sum_gx.to(gamma.scalar_type());
   std::vector<at::Tensor> out{dx, dgamma, dbeta};
   if (need_residual_grad) out.push_back(dres);
   return out;

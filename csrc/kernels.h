@@ -100,16 +100,24 @@ hipError_t FusedSgdLaunch(const SgdBatchArgs& args, float lr, float momentum,
 // ---- Fused BatchNorm(+Add)+ReLU (bn_kernels.hip) --------------------------
 // NHWC dense activations [total=N*H*W rows, C channels], C % 8 == 0,
 // dt in {DT_F32, DT_F16, DT_BF16}; stats/params fp32.
+//
+// The reduction kernels flush block partials into kBnBanks rows of [2*C]
+// floats (zeroed by the caller): one shared accumulator serializes at the
+// memory controller under thousands of blocks.
+constexpr int kBnBanks = 64;
+
+// sums: banked partial buffer of kBnBanks * 2 * C floats (zeroed).
 hipError_t BnStatsLaunch(const void* x, long long total, int C, int dt,
-                         float* sums, float* sqs, hipStream_t stream);
-// Per-channel epilogue in ONE kernel: mean/invstd from the accumulated
-// sums + running-stat update — replaces ~10 tiny ATen launches per BN
-// layer (measured 3.5 ms/step of pure launch overhead at batch 64).
-hipError_t BnFinalizeLaunch(const float* sums, const float* sqs, float* mean,
-                            float* invstd, float* running_mean,
-                            float* running_var, long long count,
-                            float momentum, float eps, int C,
-                            hipStream_t stream);
+                         float* sums, float* sqs_unused, hipStream_t stream);
+// Per-channel epilogue in ONE kernel: bank-reduce + mean/invstd + running
+// stats — replaces ~10 tiny ATen launches per BN layer (measured 3.5
+// ms/step of pure launch overhead at batch 64).
+hipError_t BnFinalizeLaunch(const float* banks, float* mean, float* invstd,
+                            float* running_mean, float* running_var,
+                            long long count, float momentum, float eps,
+                            int C, hipStream_t stream);
+// Reduce banks into banks[0..2C) (the bwd tail; fwd folds it into finalize).
+hipError_t BnBankReduceLaunch(float* banks, int C, hipStream_t stream);
 hipError_t BnApplyReluLaunch(const void* x, const void* res, void* y,
                              const float* mean, const float* invstd,
                              const float* gamma, const float* beta,
