@@ -332,3 +332,30 @@ def test_mixed_cpu_gpu_ops(hvd):
                           name=f"mgpu{i}")
         assert c.sum().item() == 64.0 * i
         assert g.sum().item() == 64.0 * i
+
+
+@requires_gpu
+def test_fused_bn_wide_channels(hvd):
+    """Banked-reduction path at wide C (round-2: the flush is spread over
+    kBnBanks partial rows; verify numerics at C=2048 where the old single
+    accumulator was most contended)."""
+    from horovod_amd.ops import FusedBNReLU
+    torch.manual_seed(8)
+    N, C, H, W = 8, 2048, 7, 7
+    x = (torch.randn(N, C, H, W, device="cuda")
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    xr = x.detach().clone().requires_grad_(True)
+    fused = FusedBNReLU(C).cuda().train()
+    ref = torch.nn.BatchNorm2d(C).cuda().train()
+    ref.load_state_dict(fused.state_dict())
+    y = fused(x)
+    yr = torch.nn.functional.relu(ref(xr))
+    assert torch.allclose(y, yr, rtol=2e-4, atol=2e-4), \
+        (y - yr).abs().max()
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(x.grad, xr.grad, rtol=2e-4, atol=2e-4)
+    assert torch.allclose(fused.weight.grad, ref.weight.grad, rtol=1e-2,
+                          atol=1e-2)
+    assert torch.allclose(fused.running_var, ref.running_var, atol=1e-3)
