@@ -62,10 +62,6 @@ int main(int argc, char** argv) {
   CHECK(hipEventCreate(&e0));
   CHECK(hipEventCreate(&e1));
 
-  struct Row {
-    const char* name;
-    double bytes;  // HBM bytes the pass must move
-  };
   auto bench = [&](const char* name, double bytes, auto fn) {
     for (int i = 0; i < 3; ++i) fn();
     CHECK(hipStreamSynchronize(s));

@@ -54,6 +54,9 @@ class _GlobalProcessSet(ProcessSet):
 global_process_set = _GlobalProcessSet()
 
 
+_known_sets = []
+
+
 def add_process_set(process_set):
     """Register a new process set.  Must be called on every rank with the
     same arguments, in the same order (collective registration)."""
@@ -62,6 +65,7 @@ def add_process_set(process_set):
     if process_set.process_set_id is not None:
         raise ValueError("process set already registered")
     process_set.process_set_id = _core.add_process_set(process_set.ranks)
+    _known_sets.append(process_set)
     # registration is collective (must be called identically on every rank,
     # reference: horovod_add_process_set operations.cc:1262-1328); barrier so
     # no rank races ahead and uses the set before peers registered it
@@ -75,4 +79,12 @@ def remove_process_set(process_set):
         return False
     _core.remove_process_set(process_set.process_set_id)
     process_set.process_set_id = None
+    if process_set in _known_sets:
+        _known_sets.remove(process_set)
     return True
+
+
+def _registered_sets():
+    """All ProcessSet objects registered in this process (global set
+    included) — used by process_set_included and tooling."""
+    return list(_known_sets)

@@ -7,7 +7,9 @@ Usage mirrors the reference (horovod/torch/__init__.py):
     opt = hvd.DistributedOptimizer(opt, named_parameters=model.named_parameters())
     hvd.broadcast_parameters(model.state_dict(), root_rank=0)
 """
-from horovod_amd.torch.compression import Compression  # noqa: F401
+from horovod_amd.torch.compression import (Compression,  # noqa: F401
+                                            Compressor, FP16Compressor,
+                                            BF16Compressor, NoneCompressor)
 from horovod_amd.torch.functions import (allgather_object,  # noqa: F401
                                          broadcast_object,
                                          broadcast_optimizer_state,
@@ -20,11 +22,15 @@ from horovod_amd.torch.mpi_ops import (  # noqa: F401
     cross_size, cuda_built, ddl_built, gloo_built, gloo_enabled,
     grouped_allgather, grouped_allgather_async, grouped_allreduce,
     grouped_allreduce_, grouped_allreduce_async, grouped_allreduce_async_,
-    grouped_reducescatter, grouped_reducescatter_async, init, is_homogeneous,
+    grouped_reducescatter, grouped_reducescatter_async,
+    handle_average_backwards_compatibility, init, is_homogeneous,
     is_initialized, join, local_rank, local_size, mpi_built, mpi_enabled,
     mpi_threads_supported, nccl_built, poll, rank, reducescatter,
-    reducescatter_async, rocm_built, shutdown, size, sparse_allreduce_async,
-    start_timeline, stop_timeline, synchronize, wait)
+    process_set_included, reducescatter_async, rocm_built, shutdown, size,
+    sparse_allreduce_async, start_timeline, stop_timeline, synchronize, wait)
+from horovod_amd.torch.mpi_ops import (  # noqa: F401  (autograd classes)
+    HorovodAllgather, HorovodAllreduce, HorovodAlltoall, HorovodBroadcast,
+    HorovodGroupedAllreduce, HorovodReducescatter)
 from horovod_amd.torch.mpi_ops import (add_process_set,  # noqa: F401
                                        remove_process_set)
 from horovod_amd.common.process_sets import (ProcessSet,  # noqa: F401

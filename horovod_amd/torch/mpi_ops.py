@@ -345,10 +345,45 @@ def _grouped_allreduce_impl(tensors, outputs, average, name, op,
                                  post_divisor=post_div)
 
 
+class HorovodGroupedAllreduce(torch.autograd.Function):
+    """Differentiable grouped allreduce (reference: mpi_ops.py
+    HorovodGroupedAllreduce): ONE fused negotiation unit; the gradient is
+    the same grouped allreduce of the incoming grads."""
+
+    @staticmethod
+    def forward(ctx, average, name, op, prescale_factor, postscale_factor,
+                process_set, *tensors):
+        ctx.average = average
+        ctx.op = op
+        ctx.prescale_factor = prescale_factor
+        ctx.postscale_factor = postscale_factor
+        ctx.process_set = process_set
+        outputs = [torch.empty_like(t) for t in tensors]
+        handle = _grouped_allreduce_impl(list(tensors), outputs, average,
+                                         name, op, prescale_factor,
+                                         postscale_factor, process_set)
+        return tuple(synchronize(handle))
+
+    @staticmethod
+    def backward(ctx, *grads):
+        gs = [g.contiguous() for g in grads]
+        outs = [torch.empty_like(g) for g in gs]
+        handle = _grouped_allreduce_impl(gs, outs, ctx.average, None, ctx.op,
+                                         ctx.prescale_factor,
+                                         ctx.postscale_factor,
+                                         ctx.process_set)
+        reduced = synchronize(handle)
+        return (None, None, None, None, None, None) + tuple(reduced)
+
+
 def grouped_allreduce(tensors, average=None, name=None, compression=None,
                       op=None, prescale_factor=1.0, postscale_factor=1.0,
                       process_set=global_process_set):
     from horovod_amd.torch.compression import Compression
+    if any(t.requires_grad for t in tensors):
+        return list(HorovodGroupedAllreduce.apply(
+            average, name, op, prescale_factor, postscale_factor,
+            process_set, *tensors))
     wire = None
     if compression is not None and compression is not Compression.none:
         wire = compression.wire_dtype(tensors[0].dtype)
@@ -443,6 +478,13 @@ def grouped_allgather_async(tensors, name=None, process_set=global_process_set):
 
 
 def grouped_allgather(tensors, name=None, process_set=global_process_set):
+    if any(t.requires_grad for t in tensors):
+        # differentiable path: per-tensor autograd allgather (reference
+        # exposes grouped gradients; the per-tensor functions negotiate
+        # concurrently through the background fusion anyway)
+        base = name or _next_name("grouped_allgather")
+        return [allgather(t, name=f"{base}.{i}", process_set=process_set)
+                for i, t in enumerate(tensors)]
     handles = grouped_allgather_async(tensors, name, process_set)
     return [synchronize(h) for h in handles]
 
@@ -636,6 +678,31 @@ def grouped_reducescatter(tensors, name=None, compression=None, op=None,
 # ---------------------------------------------------------------------------
 # join / barrier
 # ---------------------------------------------------------------------------
+def process_set_included(process_set_id=0):
+    """1 if this rank belongs to the process set, else 0 (reference:
+    operations.cc horovod_process_set_included)."""
+    from horovod_amd.common import process_sets as _ps
+    if process_set_id == 0:
+        return 1
+    for ps in _ps._registered_sets():
+        if ps.process_set_id == process_set_id:
+            return 1 if rank() in (ps.ranks or []) else 0
+    raise ValueError(f"unknown process_set_id {process_set_id}")
+
+
+def handle_average_backwards_compatibility(op, average):
+    """Map the legacy `average=` flag onto `op=` (reference:
+    horovod/common/util.py handle_average_backwards_compatibility)."""
+    if op is not None and average is not None:
+        raise ValueError("The op parameter supersedes average. Please "
+                         "provide only one of them.")
+    if op is not None:
+        return op
+    if average is not None:
+        return Average if average else Sum
+    return Average
+
+
 def join(device=-1):
     """Signal that this rank has no more data; blocks until every rank has
     joined.  Returns the last rank to join (reference: operations.cc

@@ -599,3 +599,45 @@ def test_autotune_categorical_np2(tmp_path):
     if len(rows) >= 13:
         assert any(r.split(",")[3] == "0" for r in rows), \
             "cache-off arm never explored"
+
+
+def test_grouped_allreduce_autograd_np2():
+    """Round-2: grouped allreduce is differentiable (reference
+    HorovodGroupedAllreduce) — one fused negotiation unit whose gradient
+    is the grouped allreduce of the incoming grads."""
+    run_workers(2, """
+        ts = [torch.full((8,), float(rank + 1), requires_grad=True)
+              for _ in range(3)]
+        outs = hvd.grouped_allreduce(ts, average=False, name="gar")
+        for o in outs:
+            assert torch.allclose(o, torch.full((8,), 3.0))
+        loss = sum((o * (i + 1)).sum() for i, o in enumerate(outs))
+        loss.backward()
+        # d(loss)/d(t_i) = allreduce-sum of (i+1)*ones = 2*(i+1)
+        for i, t in enumerate(ts):
+            assert torch.allclose(t.grad, torch.full((8,), 2.0 * (i + 1))), \
+                (i, t.grad)
+        # grouped allgather differentiable path
+        gs = [torch.full((2, 2), float(rank), requires_grad=True)
+              for _ in range(2)]
+        outs = hvd.grouped_allgather(gs, name="gag")
+        sum(o.sum() for o in outs).backward()
+        for g in gs:
+            assert g.grad is not None and torch.allclose(
+                g.grad, torch.full((2, 2), 2.0))
+    """)
+
+
+def test_api_surface_shims_np2():
+    run_workers(2, """
+        assert hvd.process_set_included(0) == 1
+        ps = hvd.add_process_set(hvd.ProcessSet([0]))
+        assert hvd.process_set_included(ps.process_set_id) == (
+            1 if rank == 0 else 0)
+        assert hvd.handle_average_backwards_compatibility(None, True) == \
+            hvd.Average
+        assert hvd.handle_average_backwards_compatibility(None, False) == \
+            hvd.Sum
+        assert hvd.handle_average_backwards_compatibility(hvd.Max, None) == \
+            hvd.Max
+    """)
