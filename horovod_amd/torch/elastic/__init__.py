@@ -1,4 +1,10 @@
-from horovod_amd.torch.elastic.state import TorchState  # noqa: F401
+from horovod_amd.torch.elastic.state import (TorchState,  # noqa: F401
+                                              StateHandler,
+                                              ModelStateHandler,
+                                              OptimizerStateHandler,
+                                              SamplerStateHandler,
+                                              get_handler_registry,
+                                              set_handler_registry)
 from horovod_amd.torch.elastic.sampler import ElasticSampler  # noqa: F401
 
 from horovod_amd.common.elastic import (ObjectState, State,  # noqa: F401

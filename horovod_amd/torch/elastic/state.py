@@ -14,7 +14,11 @@ from horovod_amd.torch.functions import (broadcast_object,
 from horovod_amd.torch.mpi_ops import rank
 
 
-class _Handler:
+class StateHandler:
+    """Base class for custom state handlers (reference: state.py
+    StateHandler).  Register a type with set_handler_registry/
+    get_handler_registry so TorchState picks it up for your objects."""
+
     def __init__(self, value):
         self.value = value
 
@@ -32,7 +36,7 @@ class _Handler:
         self.save()
 
 
-class ModelStateHandler(_Handler):
+class ModelStateHandler(StateHandler):
     def __init__(self, model):
         super().__init__(model)
         self._saved_model_state = copy.deepcopy(model.state_dict())
@@ -47,7 +51,7 @@ class ModelStateHandler(_Handler):
         broadcast_parameters(self.value.state_dict(), root_rank=0)
 
 
-class OptimizerStateHandler(_Handler):
+class OptimizerStateHandler(StateHandler):
     def __init__(self, optimizer):
         super().__init__(optimizer)
         self._saved_state = copy.deepcopy(optimizer.state_dict())
@@ -69,7 +73,7 @@ class OptimizerStateHandler(_Handler):
         broadcast_optimizer_state(self.value, root_rank=0)
 
 
-class SamplerStateHandler(_Handler):
+class SamplerStateHandler(StateHandler):
     def save(self):
         self._saved = self.value.state_dict()
 
@@ -85,14 +89,36 @@ class SamplerStateHandler(_Handler):
         self.value.reset()
 
 
-def _get_handler(v):
-    if isinstance(v, torch.nn.Module):
-        return ModelStateHandler(v)
-    if isinstance(v, torch.optim.Optimizer):
-        return OptimizerStateHandler(v)
+def _default_registry():
     from horovod_amd.torch.elastic.sampler import ElasticSampler
-    if isinstance(v, ElasticSampler):
-        return SamplerStateHandler(v)
+    return [(torch.nn.Module, ModelStateHandler),
+            (torch.optim.Optimizer, OptimizerStateHandler),
+            (ElasticSampler, SamplerStateHandler)]
+
+
+_handler_registry = None
+
+
+def get_handler_registry():
+    """The (type, StateHandler) pairs TorchState consults, most specific
+    first (reference: state.py get_handler_registry)."""
+    global _handler_registry
+    if _handler_registry is None:
+        _handler_registry = _default_registry()
+    return list(_handler_registry)
+
+
+def set_handler_registry(registry):
+    """Replace the handler registry (reference: set_handler_registry) —
+    e.g. to add a handler for a custom scheduler type."""
+    global _handler_registry
+    _handler_registry = list(registry)
+
+
+def _get_handler(v):
+    for typ, handler_cls in get_handler_registry():
+        if isinstance(v, typ):
+            return handler_cls(v)
     return None
 
 

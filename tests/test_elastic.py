@@ -187,3 +187,37 @@ def test_torch_state_save_restore_unit():
     for p, b in zip(model.parameters(), before):
         assert torch.allclose(p, b)
     assert state.epoch == 5
+
+
+def test_custom_state_handler_registry():
+    """Round-2: user-registered handlers (reference get/set_handler_registry)
+    are consulted by TorchState."""
+    import horovod_amd.torch.elastic as elastic
+
+    class Counter:
+        def __init__(self):
+            self.n = 0
+
+    class CounterHandler(elastic.StateHandler):
+        def save(self):
+            self._saved = self.value.n
+
+        def restore(self):
+            self.value.n = self._saved
+
+        def sync(self):
+            pass
+
+    reg = elastic.get_handler_registry()
+    elastic.set_handler_registry(reg + [(Counter, CounterHandler)])
+    try:
+        c = Counter()
+        state = elastic.TorchState(counter=c, epoch=0)
+        c.n = 5
+        state.save()
+        c.n = 9
+        state.restore()
+        assert c.n == 5
+        assert isinstance(state._handlers["counter"], CounterHandler)
+    finally:
+        elastic.set_handler_registry(reg)
