@@ -101,3 +101,22 @@ def test_elastic_cli_example(tmp_path):
          "--epochs", "1", "--samples", "256"],
         env=_env(), capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_imagenet_example_np2(tmp_path):
+    """The canonical ImageNet ResNet-50 example (synthetic mode) runs under
+    the launcher, saves a checkpoint and resumes from it."""
+    import subprocess
+    import sys
+    env = _env()
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bin", "hvdrun"), "-np", "2",
+         sys.executable,
+         os.path.join(REPO, "examples", "pytorch_imagenet_resnet50.py"),
+         "--batch-size", "2", "--steps-per-epoch", "2", "--epochs", "1",
+         "--checkpoint-format",
+         str(tmp_path / "ck-{epoch}.pt")],
+        env=env, capture_output=True, text=True, timeout=600, cwd=str(tmp_path))
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "img/s total" in r.stdout, r.stdout
+    assert (tmp_path / "ck-0.pt").exists()
