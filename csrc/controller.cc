@@ -257,6 +257,8 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
       // never cached: join/barrier are stateful.  (alltoall caches too —
       // the send-split row is part of the signature, so changed splits
       // invalidate and renegotiate.)
+      if (req.type == RequestType::JOIN)
+        my_joined_.insert(req.process_set_id);
       slow.push_back(std::move(req));
       continue;
     }
@@ -310,7 +312,12 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
         const Request& sig = cache_.GetRequest((int)s);
         auto it = process_sets_.find(sig.process_set_id);
         bool member = it != process_sets_.end() && it->second.contains(rank_);
-        vote = mine[s] || !member;
+        // a joined rank votes ready for every slot of that set — it will
+        // zero-substitute, so steady-state CACHED tensors of busier peers
+        // keep firing (reference: controller.cc:130-134 records a hit for
+        // all bits when joined; without this, join + cache fast path
+        // deadlocks)
+        vote = mine[s] || !member || my_joined_.count(sig.process_set_id);
       }
       if (vote) vecA[s / 8] |= (char)(1 << (s % 8));
     }
@@ -423,6 +430,8 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
         if (resp.type == ResponseType::JOIN || resp.type == ResponseType::BARRIER)
           inflight_.erase(SetKey(resp.process_set_id,
                                  resp.type == ResponseType::JOIN ? "join" : "barrier"));
+        if (resp.type == ResponseType::JOIN)
+          my_joined_.erase(resp.process_set_id);  // everyone joined: reset
         if (cache_enabled_) {
           int32_t li = -1;
           auto its = process_sets_.find(resp.process_set_id);

@@ -196,6 +196,9 @@ class Controller {
   // diverging fast path before it deadlocks RCCL
   bool check_stream_ = false;
   uint64_t response_hash_ = 1469598103934665603ull;
+  // process sets this rank has JOINed (votes all cache slots until the
+  // set-wide JOIN response arrives; reference controller.cc:130-134)
+  std::unordered_set<int32_t> my_joined_;
   // HOROVOD_TRACE_CYCLES=<prefix>: per-cycle debug log (prefix.<rank>)
   std::FILE* trace_ = nullptr;
   uint64_t trace_cycle_ = 0;

@@ -41,6 +41,10 @@ def main(spec_path):
             loss = loss_fn(model(x), y)
             loss.backward()
             opt.step()
+    # shards are row-group-strided, so batch counts can differ by one per
+    # rank; join() completes the trailing allreduces with zero
+    # contributions (the reference's uneven-data answer)
+    hvd.join()
 
     if hvd.rank() == 0:
         import io

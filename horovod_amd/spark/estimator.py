@@ -145,6 +145,7 @@ class TorchEstimator:
                     opt.zero_grad()
                     loss_fn(m(x), y).backward()
                     opt.step()
+            hvd.join()  # uneven shard batch counts (see _fit_task)
             if hvd.rank() == 0:
                 buf = io.BytesIO()
                 torch.save(m.state_dict(), buf)

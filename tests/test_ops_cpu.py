@@ -848,3 +848,23 @@ def test_error_mismatched_reduce_op_np2():
             assert "reduction op" in str(e).lower() or \
                 "Mismatched" in str(e), e
     """)
+
+
+def test_join_with_cached_steady_state_np2():
+    """Round-2 regression: join() while the PEER's allreduces are in the
+    cache fast path (same tensor name every step — the DistributedOptimizer
+    pattern).  A joined rank must vote ready for every cache slot
+    (reference controller.cc:130-134) or the peer's cached tensors never
+    fire and both ranks deadlock."""
+    run_workers(2, """
+        nb = 12 if rank == 0 else 5
+        for i in range(nb):
+            # SAME name each iteration -> steady-state cache fast path
+            out = hvd.allreduce(torch.ones(64), average=False, name="jcs")
+            live = 2 if i < 5 else 1
+            assert out[0].item() == float(live), (i, out[0].item())
+        hvd.join()
+        # both ranks usable afterwards
+        g = hvd.allreduce(torch.ones(4), average=False, name="after")
+        assert g[0].item() == 2.0
+    """, timeout=240)
