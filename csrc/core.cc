@@ -997,6 +997,8 @@ void PerformOperation(GlobalState& st, Response& resp) {
                 ? st.join_device[resp.process_set_id]
                 : CPU_DEVICE_ID;
     }
+    int set_n = (int)set.ranks.size();
+    int my_li = set.local_index(st.rank);
     for (size_t i = 0; i < resp.names.size(); ++i) {
       TensorTableEntry e;
       e.name = resp.names[i];
@@ -1005,10 +1007,27 @@ void PerformOperation(GlobalState& st, Response& resp) {
                       .device(dev == CPU_DEVICE_ID ? at::Device(at::kCPU)
                                                    : at::Device(at::kCUDA, dev));
       std::vector<int64_t> shape = shapes[i];
-      if ((resp.type == ResponseType::ALLGATHER ||
-           resp.type == ResponseType::ALLTOALL) &&
-          !shape.empty())
-        shape[0] = 0;  // contribute zero rows
+      // first-dim substitution must match what the RESPONSE says this rank
+      // contributes (reference: AllocateZeros(response.tensor_sizes()[i]),
+      // tensor_queue.cc:133): a CACHED response negotiated before the join
+      // still carries the old sizes, and peers size their windows by them —
+      // a hardcoded 0 here segfaults the data plane on stale-size rows
+      if (resp.type == ResponseType::ALLGATHER && !shape.empty()) {
+        int64_t rows = 0;
+        size_t per = resp.names.size()
+                         ? resp.tensor_sizes.size() / resp.names.size()
+                         : 0;
+        if (my_li >= 0 && per && (size_t)my_li < per)
+          rows = resp.tensor_sizes[i * per + my_li];
+        shape[0] = rows;
+      } else if (resp.type == ResponseType::ALLTOALL && !shape.empty()) {
+        int64_t rows = 0;
+        if (my_li >= 0 &&
+            resp.tensor_sizes.size() >= (size_t)(set_n * set_n))
+          for (int j = 0; j < set_n; ++j)
+            rows += resp.tensor_sizes[(size_t)my_li * set_n + j];
+        shape[0] = rows;
+      }
       e.tensor = at::zeros(shape, opts);
       e.device = dev;
       e.process_set_id = resp.process_set_id;

@@ -868,3 +868,19 @@ def test_join_with_cached_steady_state_np2():
         g = hvd.allreduce(torch.ones(4), average=False, name="after")
         assert g[0].item() == 2.0
     """, timeout=240)
+
+
+def test_join_after_cached_allgather_np2():
+    """Round-2 regression (segfault): a CACHED allgather slot firing while
+    ranks are joined must substitute the sizes the cached response carries
+    (reference AllocateZeros(tensor_sizes[i])), not zero rows."""
+    run_workers(2, """
+        # negotiate + cache an allgather with real contributions
+        for i in range(3):
+            g = hvd.allgather(torch.full((2, 3), float(rank)), name="cag")
+            assert g.shape == (4, 3)
+        # now both ranks join; the cached slot may fire with stale sizes
+        hvd.join()
+        out = hvd.allreduce(torch.ones(4), average=False, name="after")
+        assert out[0].item() == 2.0
+    """, timeout=240)
