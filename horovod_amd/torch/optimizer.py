@@ -135,6 +135,15 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             if self.sparse_as_dense:
                 p.grad = p.grad.to_dense()
             else:
+                # a sparse-gradient param cannot ride a dense tensor group:
+                # drop it from its group on first sighting so the group's
+                # member count matches the hooks that will actually fire
+                # (otherwise the group either never fires or fires with a
+                # sparse tensor in the fused bucket)
+                if self._groups is not None and p in self._p_to_group:
+                    gi = self._p_to_group.pop(p)
+                    self._groups[gi] = [q for q in self._groups[gi]
+                                        if q is not p]
                 return sparse_allreduce_async(
                     p.grad, name=name,
                     op=self.op if self.op != Average else Average,

@@ -641,3 +641,51 @@ def test_api_surface_shims_np2():
         assert hvd.handle_average_backwards_compatibility(hvd.Max, None) == \
             hvd.Max
     """)
+
+
+def test_kitchen_sink_training_np3():
+    """Dogfood: sparse embedding grads + dense conv/linear params with
+    gradient groups, backward_passes_per_step=2, bf16 wire compression,
+    metric averaging over a subset, and a mid-run broadcast — all in one
+    3-rank training loop that must keep ranks bit-identical."""
+    run_workers(3, """
+        torch.manual_seed(3)
+        emb = torch.nn.Embedding(50, 8, sparse=True)
+        mlp = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                  torch.nn.Linear(16, 4))
+        params = list(emb.parameters()) + list(mlp.parameters())
+        names = list(emb.named_parameters()) + list(mlp.named_parameters())
+        opt = torch.optim.SGD(params, lr=0.05)
+        opt = hvd.DistributedOptimizer(
+            opt, named_parameters=names, backward_passes_per_step=2,
+            compression=hvd.Compression.bf16, groups=2)
+        hvd.broadcast_parameters(emb.state_dict(), root_rank=0,
+                                 prefix="emb")
+        hvd.broadcast_parameters(mlp.state_dict(), root_rank=0,
+                                 prefix="mlp")
+        ps = hvd.add_process_set(hvd.ProcessSet([0, 2]))
+        from horovod_amd.torch.metrics import avg_metrics
+        for step in range(6):
+            for micro in range(2):
+                idx = torch.randint(0, 50, (16,),
+                                    generator=torch.Generator()
+                                    .manual_seed(step * 10 + micro))
+                target = torch.randn(16, 4,
+                                     generator=torch.Generator()
+                                     .manual_seed(99 + step))
+                out = mlp(emb(idx))
+                loss = torch.nn.functional.mse_loss(out, target)
+                if micro == 0:
+                    opt.zero_grad()
+                loss.backward()
+            opt.step()
+            if rank in (0, 2):
+                m = avg_metrics({"loss": float(loss)}, process_set=ps)
+                assert m["loss"] > 0
+        # every rank must hold identical weights (incl sparse embedding)
+        flat = torch.cat([p.detach().flatten() for p in params])
+        mx = hvd.allreduce(flat, op=hvd.Max, name="ksmax")
+        mn = hvd.allreduce(flat, op=hvd.Min, name="ksmin")
+        assert torch.allclose(mx, mn, atol=1e-6), \
+            (mx - mn).abs().max().item()
+    """, timeout=300)
