@@ -689,3 +689,77 @@ def test_kitchen_sink_training_np3():
         assert torch.allclose(mx, mn, atol=1e-6), \
             (mx - mn).abs().max().item()
     """, timeout=300)
+
+
+def test_sbn_model_equivalence_np2():
+    """Gold-standard dogfood: a conv+SyncBatchNorm model trained at np=2
+    (per-rank batch B, gradient averaging) must produce the same weights
+    as the identical model trained single-process on the concatenated
+    batch 2B with plain BatchNorm — step for step."""
+    run_workers(2, """
+        import torch.nn as nn
+        from horovod_amd.torch.sync_batch_norm import SyncBatchNorm
+        torch.manual_seed(11)
+        torch.use_deterministic_algorithms(True)
+
+        def make(bn):
+            torch.manual_seed(17)
+            return nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), bn(8),
+                                 nn.ReLU(), nn.Flatten(),
+                                 nn.Linear(8 * 8 * 8, 5))
+
+        dist = make(SyncBatchNorm)
+        ref = make(nn.BatchNorm2d)
+        for pd, pr in zip(dist.parameters(), ref.parameters()):
+            assert torch.equal(pd, pr)
+        opt_d = hvd.DistributedOptimizer(
+            torch.optim.SGD(dist.parameters(), lr=0.1),
+            named_parameters=dist.named_parameters())
+        opt_r = torch.optim.SGD(ref.parameters(), lr=0.1)
+
+        for step in range(4):
+            g = torch.Generator().manual_seed(200 + step)
+            full_x = torch.randn(8, 3, 8, 8, generator=g)
+            full_t = torch.randint(0, 5, (8,), generator=g)
+            my_x = full_x[rank * 4:(rank + 1) * 4]
+            my_t = full_t[rank * 4:(rank + 1) * 4]
+            opt_d.zero_grad()
+            torch.nn.functional.cross_entropy(dist(my_x), my_t).backward()
+            opt_d.step()
+            # oracle: one process, the whole batch
+            opt_r.zero_grad()
+            torch.nn.functional.cross_entropy(ref(full_x), full_t).backward()
+            opt_r.step()
+        for (n, pd), pr in zip(dist.named_parameters(), ref.parameters()):
+            assert torch.allclose(pd, pr, rtol=1e-4, atol=1e-5), \
+                (n, (pd - pr).abs().max().item())
+        # running stats must match the full-batch oracle too
+        assert torch.allclose(dist[1].running_mean, ref[1].running_mean,
+                              atol=1e-5)
+        assert torch.allclose(dist[1].running_var, ref[1].running_var,
+                              atol=1e-4)
+    """, timeout=300)
+
+
+def test_autograd_alltoall_reducescatter_np2():
+    """Gradient definitions: alltoall grad = reverse alltoall;
+    reducescatter grad = allgather of the incoming shard grads."""
+    run_workers(2, """
+        # alltoall: rank r sends rows valued 10r+dest
+        t = torch.stack([torch.full((2,), 10.0 * rank + d)
+                         for d in (0, 0, 1)]).requires_grad_(True)
+        splits = torch.tensor([2, 1]) if rank == 0 else torch.tensor([2, 1])
+        out, rs = hvd.alltoall(t, splits=splits, name="a2ag")
+        # weight received rows by (rank+1); backward reverses the exchange
+        (out.sum() * (rank + 1)).backward()
+        # my row j went to dest d(j); grad = (d(j)+1)
+        exp = torch.tensor([[1.0, 1.0], [1.0, 1.0], [2.0, 2.0]])
+        assert torch.allclose(t.grad, exp), t.grad
+        # reducescatter: grad of my shard allgathers back
+        x = torch.ones(4, 3, requires_grad=True)
+        y = hvd.reducescatter(x, op=hvd.Sum, name="rsg")
+        (y.sum() * (rank + 1)).backward()
+        # row block b of x contributes to rank b's shard; weight (b+1)
+        exp = torch.tensor([[1.0] * 3] * 2 + [[2.0] * 3] * 2)
+        assert torch.allclose(x.grad, exp), x.grad
+    """, timeout=240)
