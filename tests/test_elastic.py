@@ -221,3 +221,62 @@ def test_custom_state_handler_registry():
         assert isinstance(state._handlers["counter"], CounterHandler)
     finally:
         elastic.set_handler_registry(reg)
+
+
+def test_elastic_recovery_with_groups_and_compression(tmp_path):
+    """Elastic crash recovery while the optimizer uses tensor groups + bf16
+    wire compression (reset_distributed_state must clear group counts or
+    the respawned world deadlocks on partial groups)."""
+    worker = """
+import os, sys, time
+import torch
+import horovod_amd.torch as hvd
+import horovod_amd.torch.elastic as elastic
+
+hvd.init()
+torch.manual_seed(0)
+model = torch.nn.Sequential(torch.nn.Linear(6, 8), torch.nn.ReLU(),
+                            torch.nn.Linear(8, 2))
+opt = torch.optim.SGD(model.parameters(), lr=0.01)
+opt = hvd.DistributedOptimizer(opt,
+                               named_parameters=model.named_parameters(),
+                               compression=hvd.Compression.bf16, groups=2)
+state = elastic.TorchState(model, opt, batch=0)
+MARKER = os.environ.get("TEST_MARKER_FILE", "")
+
+@elastic.run
+def train(state):
+    while state.batch < 25:
+        if (state.batch == 6 and
+                os.environ.get("HOROVOD_WORKER_ID") == "127.0.0.1:1" and
+                not os.path.exists(MARKER + ".crashed")):
+            open(MARKER + ".crashed", "w").write("x")
+            os._exit(17)
+        time.sleep(0.02)
+        opt.zero_grad()
+        loss = model(torch.ones(3, 6)).sum()
+        loss.backward()
+        opt.step()
+        state.batch += 1
+        state.commit()
+
+train(state)
+with open(MARKER, "a") as f:
+    f.write(f"done rank={hvd.rank()} batch={state.batch}\\n")
+"""
+    from horovod_amd.runner.elastic_driver import ElasticDriver, FixedHosts
+    marker = str(tmp_path / "marker.txt")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env["TEST_MARKER_FILE"] = marker
+    env["HOROVOD_SHUTDOWN_GRACE_SECONDS"] = "2"
+    script = tmp_path / "worker.py"
+    script.write_text(worker)
+    driver = ElasticDriver(FixedHosts({"127.0.0.1": 2}),
+                           [sys.executable, str(script)], env=env,
+                           cooldown=0.0)
+    driver.start()
+    err = driver.wait_for_result(timeout=180)
+    driver.stop()
+    assert err is None, err
+    assert "batch=25" in open(marker).read()
