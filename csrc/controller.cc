@@ -98,7 +98,15 @@ int ResponseCache::AllocSlot() {
       lru = i;
     }
   }
-  if (lru >= 0) Evict(lru);
+  if (lru >= 0) {
+    Evict(lru);
+    // Evict() queues the slot on the free list for its other callers; we
+    // are handing it out RIGHT NOW — leaving it queued let the next
+    // allocation reuse the same slot under capacity pressure, aliasing two
+    // names onto one entry (fast-path deadlock: a rank votes a slot whose
+    // cached response belongs to a different tensor)
+    free_slots_.pop_back();
+  }
   return lru >= 0 ? lru : 0;
 }
 
@@ -466,9 +474,13 @@ ResponseList Controller::RunCycle(std::vector<Request> new_requests,
   }
 
   if (trace_) {
-    std::fprintf(trace_, "c%llu new=%zu slow_held=%zu pend=%zu resp=",
+    std::fprintf(trace_, "c%llu new=%zu slow_held=%zu pend=[",
                  (unsigned long long)trace_cycle_++, new_requests.size(),
-                 inflight_.size(), cached_pending_.size());
+                 inflight_.size());
+    for (auto& pr : cached_pending_)
+      std::fprintf(trace_, "%s/%d,", pr.name.c_str(),
+                   cache_.SlotOf(SetKey(pr.process_set_id, pr.name)));
+    std::fprintf(trace_, "] resp=");
     for (auto& r : result.responses) {
       std::fprintf(trace_, "%d:", (int)r.type);
       for (auto& n : r.names) std::fprintf(trace_, "%s,", n.c_str());

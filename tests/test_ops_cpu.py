@@ -884,3 +884,31 @@ def test_join_after_cached_allgather_np2():
         out = hvd.allreduce(torch.ones(4), average=False, name="after")
         assert out[0].item() == 2.0
     """, timeout=240)
+
+
+def test_cache_capacity_pressure_np3():
+    """LRU eviction churn: 40 rotating tensor names against an 8-slot cache
+    — slot allocation/eviction must stay bit-identical across ranks (the
+    fast path serializes the collective order) while values stay right."""
+    run_workers(3, """
+        for round_ in range(6):
+            for i in range(40):
+                t = torch.full((16,), float(rank + 1 + i))
+                out = hvd.allreduce(t, average=False, name=f"cap{i}")
+                assert out[0].item() == float(3 * (1 + i) + 3), (round_, i)
+    """, extra_env={"HOROVOD_CACHE_CAPACITY": "8"}, timeout=300)
+
+
+def test_many_tensor_flood_np2():
+    """300 concurrent async allreduces of mixed sizes: kernel-batch
+    splitting (48/launch), fusion-threshold splitting and handle
+    bookkeeping under load."""
+    run_workers(2, """
+        hs = []
+        for i in range(300):
+            t = torch.full((1 + (i % 97),), float(i + rank))
+            hs.append(hvd.allreduce_async(t, average=False, name=f"fl{i}"))
+        for i, h in enumerate(hs):
+            out = hvd.synchronize(h)
+            assert out[0].item() == float(2 * i + 1), (i, out[0].item())
+    """, timeout=300)
