@@ -912,3 +912,21 @@ def test_many_tensor_flood_np2():
             out = hvd.synchronize(h)
             assert out[0].item() == float(2 * i + 1), (i, out[0].item())
     """, timeout=300)
+
+
+def test_grouped_under_cache_pressure_np2():
+    """Grouped responses + LRU eviction churn: group slots must evict and
+    renegotiate coherently (partial-group cache states converge)."""
+    run_workers(2, """
+        for round_ in range(5):
+            for g in range(4):
+                ts = [torch.full((8,), float(rank + 1 + g)) for _ in range(3)]
+                outs = hvd.grouped_allreduce(ts, average=False,
+                                             name=f"grp{g}")
+                for o in outs:
+                    assert o[0].item() == float(2 * (1 + g) + 1), (round_, g)
+            for i in range(10):
+                out = hvd.allreduce(torch.ones(4), average=False,
+                                    name=f"s{i}")
+                assert out[0].item() == 2.0
+    """, extra_env={"HOROVOD_CACHE_CAPACITY": "8"}, timeout=240)
