@@ -930,3 +930,27 @@ def test_grouped_under_cache_pressure_np2():
                                     name=f"s{i}")
                 assert out[0].item() == 2.0
     """, extra_env={"HOROVOD_CACHE_CAPACITY": "8"}, timeout=240)
+
+
+def test_signature_churn_np2():
+    """Signature-invalidation churn: rotating alltoall splits, rotating
+    broadcast roots and alternating dtypes under one name must renegotiate
+    each time (INVALID path) without desync."""
+    run_workers(2, """
+        for i in range(10):
+            a, b = (1 + i % 3), (3 - i % 3)
+            t = torch.arange((a + b) * 2, dtype=torch.float32).reshape(-1, 2)
+            out, rs = hvd.alltoall(t, splits=torch.tensor([a, b]),
+                                   name="rot")
+            assert out.shape[0] == (2 * a if rank == 0 else 2 * b), i
+        for i in range(6):
+            root = i % 2
+            o = hvd.broadcast(torch.full((4,), float(rank + 1)),
+                              root_rank=root, name="rootrot")
+            assert o[0].item() == float(root + 1), i
+        for i in range(6):
+            dt = torch.float32 if i % 2 == 0 else torch.float64
+            o = hvd.allreduce(torch.ones(3, dtype=dt), average=False,
+                              name="dtrot")
+            assert o[0].item() == 2.0 and o.dtype == dt, i
+    """, timeout=240)
