@@ -466,3 +466,18 @@ def test_multi_hierarchical_allreduce():
         outs.append(out)
     for r, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0 and "HIER_OK" in out, (r, out[-500:])
+
+
+@requires_multi_gpu
+def test_multi_cache_pressure_rccl():
+    """LRU eviction churn with RCCL executing the responses: slot
+    determinism is load-bearing for the collective call order (round-2
+    slot-aliasing fix), so run the rotation at np>1 on real comms."""
+    run_workers(min(NGPU, 4), CUDA_PRELUDE + """
+        tri = size * (size + 1) // 2
+        for round_ in range(4):
+            for i in range(20):
+                t = torch.full((256,), float(rank + 1), device=dev)
+                out = hvd.allreduce(t, average=False, name=f"gcap{i}")
+                assert out[0].item() == float(tri), (round_, i)
+    """, extra_env={"HOROVOD_CACHE_CAPACITY": "8"}, timeout=420)
