@@ -481,3 +481,27 @@ def test_multi_cache_pressure_rccl():
                 out = hvd.allreduce(t, average=False, name=f"gcap{i}")
                 assert out[0].item() == float(tri), (round_, i)
     """, extra_env={"HOROVOD_CACHE_CAPACITY": "8"}, timeout=420)
+
+
+@requires_multi_gpu
+def test_multi_oneshot_with_autotune():
+    """Autotune live-adjusts fusion/cycle AND the one-shot crossover while
+    one-shot + RCCL ops interleave: eligibility must stay rank-consistent
+    through every TUNE response."""
+    run_workers(min(NGPU, 4), CUDA_PRELUDE + """
+        tri = size * (size + 1) // 2
+        import time
+        deadline = time.time() + 20
+        i = 0
+        while time.time() < deadline and i < 2000:
+            small = torch.full((2048,), float(rank + 1), device=dev)
+            o = hvd.allreduce(small, average=False, name="ta.s")
+            assert o[0].item() == float(tri), i
+            if i % 7 == 0:
+                big = torch.full((3 << 20,), 1.0, device=dev)  # 12 MB
+                ob = hvd.allreduce(big, average=False, name="ta.b")
+                assert ob[0].item() == float(size), i
+            i += 1
+    """, extra_env={"HOROVOD_ONESHOT_ALLREDUCE": "1",
+                    "HOROVOD_AUTOTUNE": "1",
+                    "HOROVOD_AUTOTUNE_WINDOW_SECONDS": "0.2"}, timeout=420)
