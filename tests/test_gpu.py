@@ -359,3 +359,23 @@ def test_fused_bn_wide_channels(hvd):
     assert torch.allclose(fused.weight.grad, ref.weight.grad, rtol=1e-2,
                           atol=1e-2)
     assert torch.allclose(fused.running_var, ref.running_var, atol=1e-3)
+
+
+@requires_gpu
+def test_per_tensor_ready_events_gpu():
+    """The HOROVOD_PER_TENSOR_READY_EVENTS=1 opt-in (fine-grained producer
+    ordering for non-default-stream users) still completes and matches."""
+    from tests.parallel_util import run_workers
+    run_workers(1, """
+        torch.cuda.set_device(0)
+        d = torch.device("cuda")
+        s = torch.cuda.Stream()
+        with torch.cuda.stream(s):
+            t = torch.ones(4096, device=d) * 3
+        out = hvd.allreduce(t, average=False, name="pte")
+        assert torch.equal(out, t)
+        ts = [torch.randn(257, device=d) for _ in range(4)]
+        outs = hvd.grouped_allreduce(ts, average=False, name="pteg")
+        for a, b in zip(ts, outs):
+            assert torch.equal(a, b)
+    """, extra_env={"HOROVOD_PER_TENSOR_READY_EVENTS": "1"}, timeout=240)
