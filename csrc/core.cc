@@ -465,7 +465,6 @@ void MeshAllreduce(GlobalState& st, const ProcessSetInfo& set,
                    const Response& resp,
                    std::vector<TensorTableEntry>& entries) {
   DataType wire = resp.dtype;
-  auto wire_t = DataTypeToTorch(wire);
   std::vector<int64_t> offsets, counts;
   PerTensorLayout(resp, offsets, counts);
   std::vector<at::Tensor> flats;
