@@ -139,6 +139,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                  weight_decay, dampening, nesterov);
         });
 
+  m.def("fused_adamw_step",
+        [](std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+           std::vector<at::Tensor> exp_avgs,
+           std::vector<at::Tensor> exp_avg_sqs, double lr, double beta1,
+           double beta2, double eps, double weight_decay, int64_t step) {
+          hvd::gpu::FusedAdamwStep(params, grads, exp_avgs, exp_avg_sqs, lr,
+                                   beta1, beta2, eps, weight_decay, step);
+        });
+
   m.def("fused_bn_relu_forward",
         [](at::Tensor x, py::object residual, at::Tensor gamma, at::Tensor beta,
            at::Tensor rm, at::Tensor rv, double momentum, double eps) {

@@ -1564,6 +1564,36 @@ void FusedSgdStep(std::vector<at::Tensor>& params,
   flush();
 }
 
+void FusedAdamwStep(std::vector<at::Tensor>& params,
+                    std::vector<at::Tensor>& grads,
+                    std::vector<at::Tensor>& exp_avgs,
+                    std::vector<at::Tensor>& exp_avg_sqs, double lr,
+                    double beta1, double beta2, double eps,
+                    double weight_decay, int64_t step) {
+  if (params.empty()) return;
+  int device = (int)params[0].get_device();
+  c10::hip::HIPGuard guard(device);
+  hipStream_t stream = c10::hip::getCurrentHIPStream(device).stream();
+  AdamwBatchArgs args;
+  auto flush = [&] {
+    if (args.count == 0) return;
+    HIP_CHECK(FusedAdamwLaunch(args, (float)lr, (float)beta1, (float)beta2,
+                               (float)eps, (float)weight_decay, step,
+                               stream));
+    args.count = 0;
+  };
+  for (size_t i = 0; i < params.size(); ++i) {
+    if (args.count == kCopyBatchCapacity) flush();
+    int k = args.count++;
+    args.params[k] = params[i].data_ptr();
+    args.grads[k] = grads[i].data_ptr();
+    args.exp_avg[k] = exp_avgs[i].data_ptr();
+    args.exp_avg_sq[k] = exp_avg_sqs[i].data_ptr();
+    args.numel[k] = (unsigned long long)params[i].numel();
+  }
+  flush();
+}
+
 // ---- Fused BN(+Add)+ReLU host side ----------------------------------------
 // Returns {y, save_mean, save_invstd}.  Small per-channel math ([C] tensors)
 // runs through ATen; the big passes are the CDNA4 kernels.

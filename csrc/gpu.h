@@ -73,5 +73,14 @@ void FusedSgdStep(std::vector<at::Tensor>& params,
                   std::vector<at::Tensor>& momenta, double lr, double momentum,
                   double weight_decay, double dampening, bool nesterov);
 
+// Fused AdamW step (decoupled weight decay, torch.optim.AdamW math);
+// step is the 1-based step count for bias correction.  fp32 tensors.
+void FusedAdamwStep(std::vector<at::Tensor>& params,
+                    std::vector<at::Tensor>& grads,
+                    std::vector<at::Tensor>& exp_avgs,
+                    std::vector<at::Tensor>& exp_avg_sqs, double lr,
+                    double beta1, double beta2, double eps,
+                    double weight_decay, int64_t step);
+
 }  // namespace gpu
 }  // namespace hvd

@@ -97,6 +97,21 @@ hipError_t FusedSgdLaunch(const SgdBatchArgs& args, float lr, float momentum,
                           float weight_decay, float dampening, bool nesterov,
                           hipStream_t stream);
 
+// ---- Fused AdamW (adamw_kernels.hip) --------------------------------------
+struct AdamwBatchArgs {
+  void* params[kCopyBatchCapacity];
+  const void* grads[kCopyBatchCapacity];
+  void* exp_avg[kCopyBatchCapacity];
+  void* exp_avg_sq[kCopyBatchCapacity];
+  unsigned long long numel[kCopyBatchCapacity];
+  int count = 0;
+};
+
+// step is 1-based (bias correction uses beta^step).
+hipError_t FusedAdamwLaunch(const AdamwBatchArgs& args, float lr, float beta1,
+                            float beta2, float eps, float weight_decay,
+                            long long step, hipStream_t stream);
+
 // ---- Fused BatchNorm(+Add)+ReLU (bn_kernels.hip) --------------------------
 // NHWC dense activations [total=N*H*W rows, C channels], C % 8 == 0,
 // dt in {DT_F32, DT_F16, DT_BF16}; stats/params fp32.
