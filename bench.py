@@ -37,6 +37,9 @@ def parse_args():
     p.add_argument("--use-adasum", action="store_true")
     p.add_argument("--no-bf16", action="store_true",
                    help="disable bf16 autocast (fp32 compute)")
+    p.add_argument("--optimizer", default="sgd",
+                   choices=["sgd", "adamw"],
+                   help="adamw uses the fused CDNA4 AdamW kernel")
     p.add_argument("--fused-sgd", dest="fused_sgd", action="store_true",
                    default=True,
                    help="use the CDNA4 fused SGD step kernel (default)")
@@ -98,7 +101,11 @@ def main():
     compression = {"none": Compression.none, "fp16": Compression.fp16,
                    "bf16": Compression.bf16}[args.compression]
 
-    if args.fused_sgd:
+    if args.optimizer == "adamw":
+        from horovod_amd.ops import FusedAdamW
+        opt = FusedAdamW(model.parameters(), lr=1e-4 * hvd.size(),
+                         weight_decay=0.01)
+    elif args.fused_sgd:
         from horovod_amd.ops import FusedSGD
         opt = FusedSGD(model.parameters(), lr=0.0125 * hvd.size(),
                        momentum=0.9, weight_decay=5e-5)
@@ -217,6 +224,7 @@ def main():
                 "parallelism": f"dp{n}",
                 "compression": args.compression,
                 "reduction": "adasum" if args.use_adasum else "average",
+                "optimizer": args.optimizer,
                 "fused_sgd": args.fused_sgd,
                 "fused_bn": args.fused_bn,
                 "hipgraph": args.hipgraph,
