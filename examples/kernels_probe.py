@@ -59,4 +59,13 @@ for p in params:
 opt = FusedSGD(params, lr=0.01, momentum=0.9, weight_decay=1e-4)
 timed("fused SGD step, 160 params (~25M elems)", opt.step,
       bytes_moved=5 * 4 * sum(p.numel() for p in params))
+
+# 5) fused AdamW, BERT-shaped params (p,g,m,v read + p,m,v write = 7 passes)
+from horovod_amd.ops import FusedAdamW
+aparams = [torch.randn(n, device="cuda") for n in [340_000_000 // 400] * 64]
+for p in aparams:
+    p.grad = torch.randn_like(p)
+aopt = FusedAdamW(aparams, lr=1e-4)
+timed("fused AdamW step, 64 params (~54M elems)", aopt.step,
+      bytes_moved=7 * 4 * sum(p.numel() for p in aparams))
 hvd.shutdown()
