@@ -24,8 +24,7 @@ setup(
               "horovod_amd.torch.elastic", "horovod_amd.runner",
               "horovod_amd.data", "horovod_amd.models", "horovod_amd.ops",
               "horovod_amd.parallel", "horovod_amd.utils",
-              "horovod_amd.spark", "horovod_amd.ray",
-              "horovod_amd.elastic"],
+              "horovod_amd.spark", "horovod_amd.ray"],
     ext_modules=[Extension("horovod_amd._core", sources=[])],
     cmdclass={"build_ext": HipBuildExt},
     scripts=["bin/hvdrun", "bin/horovodrun"],
