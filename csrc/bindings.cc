@@ -170,7 +170,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     SetTimeline(st, std::make_shared<Timeline>(path, st.rank));
     (void)mark_cycles;
   });
-  m.def("stop_timeline", [] { SetTimeline(State(), nullptr); });
+  m.def("stop_timeline", [] {
+    auto tl = GetTimeline(State());
+    SetTimeline(State(), nullptr);
+    if (tl) {
+      py::gil_scoped_release nogil;
+      tl->Finalize();  // file is complete, valid JSON on return
+    }
+  });
 
   // ---- process sets -------------------------------------------------------
   m.def("add_process_set", [](std::vector<int32_t> ranks) {

@@ -38,23 +38,31 @@ Timeline::Timeline(const std::string& path, int rank)
   writer_ = std::thread([this] { WriterLoop(); });
 }
 
-Timeline::~Timeline() {
+Timeline::~Timeline() { Finalize(); }
+
+void Timeline::Finalize() {
+  bool expected = false;
+  if (!finalized_.compare_exchange_strong(expected, true)) return;
   stop_ = true;
   cv_.notify_all();
   if (writer_.joinable()) writer_.join();
+  std::lock_guard<std::mutex> g(mu_);
   if (file_) {
-    // keep the file valid JSON (reference: seek-back patching; we simply
-    // close the array).
+    // close the JSON array so the file is valid the moment stop returns
+    // (reference keeps validity by seek-back patching after every write)
     std::fputs("\n]\n", file_);
     std::fclose(file_);
+    file_ = nullptr;
   }
 }
 
 int Timeline::PidOf(const std::string& tensor) {
+  std::unique_lock<std::mutex> lk(pid_mu_);
   auto it = pids_.find(tensor);
   if (it != pids_.end()) return it->second;
   int pid = next_pid_++;
   pids_[tensor] = pid;
+  lk.unlock();
   char buf[512];
   std::snprintf(buf, sizeof(buf),
                 "{\"ph\":\"M\",\"name\":\"process_name\",\"pid\":%d,\"args\":{"
@@ -70,6 +78,7 @@ int Timeline::PidOf(const std::string& tensor) {
 }
 
 void Timeline::Push(std::string json) {
+  if (finalized_) return;  // late event from a still-held reference
   std::lock_guard<std::mutex> g(mu_);
   queue_.push_back({std::move(json)});
   cv_.notify_one();

@@ -35,6 +35,13 @@ class Timeline {
   void NegotiateEnd(const std::string& tensor);
   void Marker(const std::string& name);
 
+  // Drain the writer, close the JSON array and the file.  SYNCHRONOUS and
+  // idempotent: hvd.stop_timeline() calls this so the file is valid,
+  // complete JSON the moment the call returns (late events from references
+  // still held by other threads become no-ops).
+
+  void Finalize();
+
   int64_t NowUs() const {
     return std::chrono::duration_cast<std::chrono::microseconds>(
                std::chrono::steady_clock::now() - t0_)
@@ -54,9 +61,13 @@ class Timeline {
   std::mutex mu_;
   std::condition_variable cv_;
   std::deque<Record> queue_;
+  // pids_ is touched from the background thread AND the GPU finalizer
+  // thread (Activity) — needs its own lock
+  std::mutex pid_mu_;
   std::unordered_map<std::string, int> pids_;
   int next_pid_ = 1;
   std::atomic<bool> stop_{false};
+  std::atomic<bool> finalized_{false};
   bool first_ = true;
   std::thread writer_;
 };

@@ -795,3 +795,25 @@ def test_expert_parallel_routing_np4():
         # expert grads exist and are LOCAL (per-rank expert => not averaged)
         assert expert.weight.grad is not None
     """, timeout=300)
+
+
+def test_timeline_stop_is_synchronous_np2():
+    """Round-2 regression: stop_timeline() must return with the file
+    already complete, valid JSON (previously the writer finalized
+    asynchronously, so an immediate read raced the flush and one rank's
+    crash cascaded into peer connection resets)."""
+    import json as _json
+    run_workers(2, """
+        import json
+        for cycle in range(8):
+            base = f"/tmp/hvd_tl_sync_{rank}_{cycle}.json"
+            hvd.start_timeline(base if rank == 0 else base[:-5])
+            for i in range(8):
+                hvd.allreduce(torch.ones(128), average=False, name=f"t{i}")
+            hvd.stop_timeline()
+            mine = base if rank == 0 else base[:-5] + ".1"
+            data = json.load(open(mine))
+            assert isinstance(data, list) and data, (cycle, mine)
+            import os
+            os.unlink(mine)
+    """, timeout=240)
