@@ -954,3 +954,32 @@ def test_signature_churn_np2():
                               name="dtrot")
             assert o[0].item() == 2.0 and o.dtype == dt, i
     """, timeout=240)
+
+
+def test_mixed_barrier_subset_interleaving_np4():
+    """Randomized (rank-identical) interleaving of barriers, global
+    allreduces and three overlapping subset allreduces at np=4."""
+    run_workers(4, """
+        import random
+        rng = random.Random(42)
+        sets = [hvd.add_process_set(hvd.ProcessSet([0, 1])),
+                hvd.add_process_set(hvd.ProcessSet([2, 3])),
+                hvd.add_process_set(hvd.ProcessSet([0, 2, 3]))]
+        members = [[0, 1], [2, 3], [0, 2, 3]]
+        for i in range(30):
+            k = rng.randint(0, 3)
+            if k == 0:
+                hvd.barrier()
+            elif k == 1:
+                out = hvd.allreduce(torch.ones(32) * (rank + 1),
+                                    average=False, name=f"g{i}")
+                assert out[0].item() == 10.0
+            else:
+                si = rng.randint(0, 2)
+                if rank in members[si]:
+                    out = hvd.allreduce(torch.ones(8), average=False,
+                                        name=f"s{si}.{i}",
+                                        process_set=sets[si])
+                    assert out[0].item() == float(len(members[si]))
+        hvd.barrier()
+    """, timeout=240)
