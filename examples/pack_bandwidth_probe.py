@@ -5,6 +5,9 @@ ncclAllReduce(self) -> unpack) and the raw batched-copy kernel via a
 grouped allreduce of many tensors.  Run under rocprofv3 for per-kernel
 numbers; committed summaries live in profiles/.
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import time
 
 import torch

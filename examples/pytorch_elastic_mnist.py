@@ -4,6 +4,9 @@ pytorch_mnist_elastic.py):
     hvdrun --host-discovery-script ./discover_hosts.sh \
         --min-np 1 --max-np 8 python examples/pytorch_elastic_mnist.py
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import argparse
 
 import torch

@@ -4,6 +4,9 @@ Runs on synthetic MNIST-shaped data (no network access for datasets):
 
     hvdrun -np 2 python examples/pytorch_mnist.py --epochs 2
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import argparse
 
 import torch

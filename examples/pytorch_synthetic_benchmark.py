@@ -4,6 +4,9 @@ examples/pytorch/pytorch_synthetic_benchmark.py flag surface (13-35):
     hvdrun -np 8 python examples/pytorch_synthetic_benchmark.py \
         --model resnet50 --batch-size 64 [--fp16-allreduce] [--use-adasum]
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import argparse
 import timeit
 
